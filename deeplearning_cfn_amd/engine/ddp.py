@@ -1,0 +1,271 @@
+"""L4 — the data-parallel engine.
+
+Re-realises both reference examples' training semantics (SURVEY.md §3.3
+scope ruling) in one engine:
+
+* the Horovod/NCCL path — ``DistributedOptimizer`` gradient hooks fusing
+  grads into ~64 MB buffers reduced by ``ncclAllReduce`` overlapped with
+  backward — becomes backward-hook-driven bucketed
+  ``all_reduce`` on RCCL over xGMI;
+* the MXNet ``dist_device_sync`` parameter-server path
+  maps onto the same synchronous device-aggregated all-reduce (push+pull
+  of a layer == its slice of a reduced bucket); ``dist_async`` is kept as
+  a one-step-stale variant.
+
+MI355X-first decisions:
+
+* **Flat buckets, gradient-as-bucket-view.**  Each bucket is one
+  contiguous tensor; every ``param.grad`` is a view into it.  Backward
+  accumulates straight into the flat buffer, the collective reduces it
+  without packing copies, and the fused-SGD HIP kernel (ops/) updates the
+  whole flat param/grad pair in one launch.
+* **Bucket count sized for the xGMI mesh**: 8 GPUs are a full mesh of
+  7 point-to-point links per GPU (~153 GB/s each); a single in-flight
+  collective rides ~1 link per direction.  Default: aim for >=
+  ``XGMI_LINKS_PER_GPU`` buckets in flight (bucket bytes ~ total/7,
+  clamped to [1 MiB, 64 MiB]) so RCCL's channels keep the mesh fed, and
+  the first (deepest) buckets start reducing while backward still runs.
+* Collectives are issued ``async_op=True`` from the post-accumulate-grad
+  hook; torch's NCCL process group runs them on its own HIP stream, so
+  comm/compute overlap needs no manual stream juggling; ``finalize()``
+  makes the compute stream wait and applies the 1/N scaling.
+"""
+
+from __future__ import annotations
+
+import contextlib
+from dataclasses import dataclass, field
+from typing import Dict, List, Optional
+
+import torch
+import torch.distributed as dist
+
+from ..topo import XGMI_LINKS_PER_GPU
+
+_MIN_BUCKET = 1 << 20          # 1 MiB
+_MAX_BUCKET = 64 << 20         # 64 MiB (Horovod's fusion default)
+
+
+@dataclass
+class Bucket:
+    index: int
+    params: List[torch.nn.Parameter]
+    flat: torch.Tensor
+    views: List[torch.Tensor]
+    pending: int = 0
+    work: Optional[object] = None  # dist.Work
+    # dist_async double-buffer: alternate send buffers so step t's
+    # collective runs while step t consumes step t-1's result.
+    async_bufs: Optional[List[torch.Tensor]] = None
+    async_works: Optional[List[Optional[object]]] = None
+    async_idx: int = 0
+
+
+@dataclass
+class BucketPlan:
+    buckets: List[Bucket] = field(default_factory=list)
+
+    @property
+    def nbytes(self) -> List[int]:
+        return [b.flat.numel() * b.flat.element_size() for b in self.buckets]
+
+
+def _auto_bucket_bytes(total_bytes: int) -> int:
+    """total/links, clamped — keeps ~7 collectives in flight."""
+    return max(_MIN_BUCKET, min(_MAX_BUCKET, total_bytes // XGMI_LINKS_PER_GPU))
+
+
+class DataParallel:
+    """Wraps a module with bucketed-all-reduce data parallelism.
+
+    Unlike ``torch.nn.parallel.DistributedDataParallel`` this engine owns
+    its buckets as flat buffers exposed to the fused optimizer, implements
+    the kvstore-mode semantics of the MXNet example, and works with
+    world_size == 1 (no process group) for single-GPU runs.
+    """
+
+    def __init__(
+        self,
+        module: torch.nn.Module,
+        bucket_bytes: Optional[int] = None,   # None => xGMI-auto
+        mode: str = "dist_device_sync",       # | dist_sync | dist_async | local
+        broadcast_params: bool = True,
+        process_group: Optional[object] = None,
+    ):
+        if mode not in ("dist_device_sync", "dist_sync", "dist_async", "local"):
+            raise ValueError(f"unknown kvstore mode {mode!r}")
+        self.module = module
+        self.mode = mode
+        self.pg = process_group
+        self.world_size = dist.get_world_size(self.pg) if dist.is_initialized() else 1
+        self._sync_enabled = True
+        self._hooks = []
+
+        params = [p for p in module.parameters() if p.requires_grad]
+        total = sum(p.numel() * p.element_size() for p in params)
+        self.bucket_bytes = bucket_bytes or _auto_bucket_bytes(total)
+        self.plan = self._build_plan(params)
+        self._param_bucket: Dict[int, Bucket] = {}
+        for b in self.plan.buckets:
+            for p in b.params:
+                self._param_bucket[id(p)] = b
+
+        if broadcast_params and self.world_size > 1:
+            self.broadcast_parameters()
+        self._install_hooks()
+
+    # -- construction -----------------------------------------------------
+
+    def _build_plan(self, params: List[torch.nn.Parameter]) -> BucketPlan:
+        """Group params into flat buckets in reverse creation order.
+
+        Reverse order approximates gradient-ready order during backward
+        (outputs first), so the first bucket to fill is the first to
+        reduce — the overlap window the Horovod design exploits.
+        Buckets never mix dtypes or devices.
+        """
+        plan = BucketPlan()
+        cur: List[torch.nn.Parameter] = []
+        cur_bytes = 0
+        key = None
+
+        def flush():
+            nonlocal cur, cur_bytes
+            if not cur:
+                return
+            n = sum(p.numel() for p in cur)
+            flat = torch.zeros(
+                n, dtype=cur[0].dtype, device=cur[0].device)
+            views, off = [], 0
+            for p in cur:
+                v = flat[off:off + p.numel()].view_as(p)
+                views.append(v)
+                off += p.numel()
+            b = Bucket(index=len(plan.buckets), params=list(cur),
+                       flat=flat, views=views)
+            # gradient-as-bucket-view: grads accumulate into the flat buffer
+            for p, v in zip(b.params, b.views):
+                p.grad = v
+            plan.buckets.append(b)
+            cur, cur_bytes = [], 0
+
+        for p in reversed(params):
+            k = (p.dtype, p.device)
+            if key is not None and (k != key or cur_bytes >= self.bucket_bytes):
+                flush()
+            key = k
+            cur.append(p)
+            cur_bytes += p.numel() * p.element_size()
+        flush()
+        return plan
+
+    def _install_hooks(self) -> None:
+        for b in self.plan.buckets:
+            for p in b.params:
+                h = p.register_post_accumulate_grad_hook(self._on_grad_ready)
+                self._hooks.append(h)
+
+    # -- collective machinery ---------------------------------------------
+
+    def _on_grad_ready(self, p: torch.nn.Parameter) -> None:
+        if self.world_size == 1 or self.mode == "local" or not self._sync_enabled:
+            return
+        b = self._param_bucket[id(p)]
+        b.pending -= 1
+        if b.pending == 0 and self.mode != "dist_async":
+            b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
+                                     group=self.pg, async_op=True)
+
+    def zero_grad(self) -> None:
+        """Arm the buckets for the next backward (and zero flat buffers)."""
+        for b in self.plan.buckets:
+            b.flat.zero_()
+            b.pending = len(b.params)
+            b.work = None
+
+    def forward(self, *args, **kwargs):
+        return self.module(*args, **kwargs)
+
+    __call__ = forward
+
+    @contextlib.contextmanager
+    def no_sync(self):
+        """Gradient accumulation: skip the collective this backward."""
+        prev = self._sync_enabled
+        self._sync_enabled = False
+        try:
+            yield
+        finally:
+            self._sync_enabled = prev
+
+    def finalize_backward(self) -> None:
+        """Complete the reduction before the optimizer step.
+
+        dist_device_sync / dist_sync: wait for this step's collectives
+        (the compute stream is made to wait on RCCL's stream) and scale by
+        1/N.  dist_async: do not wait for this step's works — wait for the
+        PREVIOUS step's (one-step-stale gradients, the documented MXNet
+        ``dist_async`` semantics honestly collapsed to DP).
+        """
+        if self.world_size == 1 or self.mode == "local" or not self._sync_enabled:
+            return
+        inv = 1.0 / self.world_size
+        if self.mode == "dist_async":
+            # One-step-stale gradient exchange: send this step's grads
+            # asynchronously, consume last step's average.  First step
+            # proceeds on local gradients (the PS "pull before first push"
+            # startup, SURVEY.md §4.2, honestly mapped).
+            for b in self.plan.buckets:
+                if b.async_bufs is None:
+                    b.async_bufs = [torch.empty_like(b.flat) for _ in range(2)]
+                    b.async_works = [None, None]
+                i = b.async_idx
+                prev = i ^ 1
+                prev_work = b.async_works[prev]
+                b.async_works[i] = None
+                send = b.async_bufs[i]
+                send.copy_(b.flat)
+                b.async_works[i] = dist.all_reduce(
+                    send, op=dist.ReduceOp.SUM, group=self.pg, async_op=True)
+                if prev_work is not None:
+                    prev_work.wait()
+                    b.flat.copy_(b.async_bufs[prev]).mul_(inv)
+                    b.async_works[prev] = None
+                b.async_idx = prev
+            return
+        for b in self.plan.buckets:
+            if b.pending != 0 and b.work is None:
+                # params that never got grads this step (unused branches):
+                # reduce now so ranks stay in lockstep.
+                b.work = dist.all_reduce(b.flat, op=dist.ReduceOp.SUM,
+                                         group=self.pg, async_op=True)
+            if b.work is not None:
+                b.work.wait()
+                b.flat.mul_(inv)
+                b.work = None
+
+    # -- parity utilities --------------------------------------------------
+
+    def broadcast_parameters(self, src: int = 0) -> None:
+        """``hvd.broadcast_global_variables(0)`` parity."""
+        if self.world_size == 1:
+            return
+        with torch.no_grad():
+            for p in self.module.parameters():
+                dist.broadcast(p.data, src=src, group=self.pg)
+            for bname, buf in self.module.named_buffers():
+                dist.broadcast(buf, src=src, group=self.pg)
+
+    def flat_pairs(self):
+        """(flat_param_like, flat_grad) pairs for the fused optimizer.
+
+        Yields each bucket's params and its flat grad buffer; the fused
+        SGD kernel walks the params' storage through the same offsets.
+        """
+        for b in self.plan.buckets:
+            yield b.params, b.flat
+
+    def detach_hooks(self) -> None:
+        for h in self._hooks:
+            h.remove()
+        self._hooks.clear()

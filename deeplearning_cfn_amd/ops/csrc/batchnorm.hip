@@ -1,0 +1,447 @@
+// NHWC BatchNorm (training fwd / eval fwd / bwd), fused with ReLU.
+//
+// Replaces cuDNN BatchNorm + the adjacent framework ReLU (SURVEY.md §3.2
+// rows "cuDNN BatchNorm", "framework elementwise").  Design:
+//  * x is channels-last ([R=N*H*W, C] with C contiguous) — the native
+//    layout of the conv path; statistics accumulate in fp32;
+//  * reduce kernel: each block owns a (channel-tile, row-stripe); LDS
+//    per-channel partials, one global fp32 atomicAdd per channel/block;
+//  * normalize / dx kernels: vectorized 16 B/lane when C % width == 0;
+//  * ReLU is fused: fwd applies max(0,·); bwd masks dy by (y > 0) before
+//    the reductions (y is the saved post-activation output).
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include "dlx_common.h"
+
+namespace dlx {
+
+namespace {
+
+// channel-tile width: largest power of two <= 256 dividing C
+static int channel_tile(int C) {
+  int ct = 256;
+  while (ct > 1 && (C % ct)) ct >>= 1;
+  return ct;
+}
+
+// ---- reduction: sum(x), sum(x^2) per channel ---------------------------
+
+template <typename T>
+__global__ void bn_reduce_fwd_kernel(const T* __restrict__ x,
+                                     float* __restrict__ sums,  // [2C]
+                                     long R, int C, int CT) {
+  extern __shared__ float lds[];          // [2*CT]
+  float* lsum = lds;
+  float* lsq = lds + CT;
+  for (int i = threadIdx.x; i < 2 * CT; i += blockDim.x) lds[i] = 0.f;
+  __syncthreads();
+
+  const int c = blockIdx.y * CT + (threadIdx.x % CT);
+  const int sub = threadIdx.x / CT;
+  const int rows_grp = blockDim.x / CT;
+  float s = 0.f, sq = 0.f;
+  for (long r = (long)blockIdx.x * rows_grp + sub; r < R;
+       r += (long)gridDim.x * rows_grp) {
+    float v = VecTraits<T>::load1(x + r * C + c);
+    s += v;
+    sq += v * v;
+  }
+  atomicAdd(&lsum[threadIdx.x % CT], s);
+  atomicAdd(&lsq[threadIdx.x % CT], sq);
+  __syncthreads();
+  if (threadIdx.x < CT) {
+    atomicAdd(&sums[blockIdx.y * CT + threadIdx.x], lsum[threadIdx.x]);
+    atomicAdd(&sums[C + blockIdx.y * CT + threadIdx.x], lsq[threadIdx.x]);
+  }
+}
+
+// ---- finalize: stats + running update + scale/shift --------------------
+
+__global__ void bn_finalize_fwd_kernel(const float* __restrict__ sums,
+                                       const float* __restrict__ w,
+                                       const float* __restrict__ b,
+                                       float* __restrict__ rmean,
+                                       float* __restrict__ rvar,
+                                       float* __restrict__ save_mean,
+                                       float* __restrict__ save_invstd,
+                                       float* __restrict__ scale,
+                                       float* __restrict__ shift,
+                                       long R, int C, float momentum,
+                                       float eps) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float mean = sums[c] / R;
+  float var = fmaxf(sums[C + c] / R - mean * mean, 0.f);  // biased
+  float invstd = rsqrtf(var + eps);
+  save_mean[c] = mean;
+  save_invstd[c] = invstd;
+  // torch semantics: running stats use the UNBIASED variance
+  float unbiased = (R > 1) ? var * (float)R / (float)(R - 1) : var;
+  rmean[c] = (1.f - momentum) * rmean[c] + momentum * mean;
+  rvar[c] = (1.f - momentum) * rvar[c] + momentum * unbiased;
+  float sc = w[c] * invstd;
+  scale[c] = sc;
+  shift[c] = b[c] - mean * sc;
+}
+
+__global__ void bn_eval_coeffs_kernel(const float* __restrict__ w,
+                                      const float* __restrict__ b,
+                                      const float* __restrict__ rmean,
+                                      const float* __restrict__ rvar,
+                                      float* __restrict__ scale,
+                                      float* __restrict__ shift,
+                                      int C, float eps) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  float invstd = rsqrtf(rvar[c] + eps);
+  float sc = w[c] * invstd;
+  scale[c] = sc;
+  shift[c] = b[c] - rmean[c] * sc;
+}
+
+// ---- normalize: y = x*scale[c] + shift[c] (+ReLU) ----------------------
+
+template <typename T, bool RELU, bool VEC>
+__global__ void bn_norm_kernel(const T* __restrict__ x, T* __restrict__ y,
+                               const float* __restrict__ scale,
+                               const float* __restrict__ shift,
+                               long total, int C) {
+  using VT = VecTraits<T>;
+  using Vec = typename VT::Vec;
+  const long stride = (long)gridDim.x * blockDim.x;
+  if (VEC) {
+    const long nvec = total / VT::width;
+    for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < nvec;
+         i += stride) {
+      long base = i * VT::width;
+      int c0 = (int)(base % C);
+      Vec xv = reinterpret_cast<const Vec*>(x)[i];
+      Vec yv;
+#pragma unroll
+      for (int j = 0; j < VT::width; ++j) {
+        float v = VT::load(xv, j) * scale[c0 + j] + shift[c0 + j];
+        VT::store(yv, j, RELU ? fmaxf(v, 0.f) : v);
+      }
+      reinterpret_cast<Vec*>(y)[i] = yv;
+    }
+  } else {
+    for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < total;
+         i += stride) {
+      int c = (int)(i % C);
+      float v = VT::load1(x + i) * scale[c] + shift[c];
+      VT::store1(y + i, RELU ? fmaxf(v, 0.f) : v);
+    }
+  }
+}
+
+// ---- backward reduce: sum(dy), sum(dy * xhat) per channel --------------
+
+template <typename T, bool RELU>
+__global__ void bn_reduce_bwd_kernel(const T* __restrict__ dy,
+                                     const T* __restrict__ x,
+                                     const T* __restrict__ y,
+                                     const float* __restrict__ save_mean,
+                                     const float* __restrict__ save_invstd,
+                                     float* __restrict__ sums,  // [2C]
+                                     long R, int C, int CT) {
+  extern __shared__ float lds[];
+  float* s_dy = lds;
+  float* s_dyx = lds + CT;
+  for (int i = threadIdx.x; i < 2 * CT; i += blockDim.x) lds[i] = 0.f;
+  __syncthreads();
+
+  const int c = blockIdx.y * CT + (threadIdx.x % CT);
+  const int sub = threadIdx.x / CT;
+  const int rows_grp = blockDim.x / CT;
+  const float mean = save_mean[c], invstd = save_invstd[c];
+  float a = 0.f, bx = 0.f;
+  for (long r = (long)blockIdx.x * rows_grp + sub; r < R;
+       r += (long)gridDim.x * rows_grp) {
+    long i = r * C + c;
+    float g = VecTraits<T>::load1(dy + i);
+    if (RELU && VecTraits<T>::load1(y + i) <= 0.f) g = 0.f;
+    float xh = (VecTraits<T>::load1(x + i) - mean) * invstd;
+    a += g;
+    bx += g * xh;
+  }
+  atomicAdd(&s_dy[threadIdx.x % CT], a);
+  atomicAdd(&s_dyx[threadIdx.x % CT], bx);
+  __syncthreads();
+  if (threadIdx.x < CT) {
+    atomicAdd(&sums[blockIdx.y * CT + threadIdx.x], s_dy[threadIdx.x]);
+    atomicAdd(&sums[C + blockIdx.y * CT + threadIdx.x], s_dyx[threadIdx.x]);
+  }
+}
+
+__global__ void bn_finalize_bwd_kernel(const float* __restrict__ sums,
+                                       const float* __restrict__ w,
+                                       const float* __restrict__ save_invstd,
+                                       float* __restrict__ dgamma,
+                                       float* __restrict__ dbeta,
+                                       float* __restrict__ k1,  // sum_dy/R
+                                       float* __restrict__ k2,  // sum_dyxhat/R
+                                       float* __restrict__ gi,  // gamma*invstd
+                                       long R, int C) {
+  int c = blockIdx.x * blockDim.x + threadIdx.x;
+  if (c >= C) return;
+  dbeta[c] = sums[c];
+  dgamma[c] = sums[C + c];
+  k1[c] = sums[c] / R;
+  k2[c] = sums[C + c] / R;
+  gi[c] = w[c] * save_invstd[c];
+}
+
+// dx = gi[c] * (dy_eff - k1[c] - xhat * k2[c])
+template <typename T, bool RELU, bool VEC>
+__global__ void bn_dx_kernel(const T* __restrict__ dy,
+                             const T* __restrict__ x,
+                             const T* __restrict__ y,
+                             const float* __restrict__ save_mean,
+                             const float* __restrict__ save_invstd,
+                             const float* __restrict__ k1,
+                             const float* __restrict__ k2,
+                             const float* __restrict__ gi,
+                             T* __restrict__ dx, long total, int C) {
+  using VT = VecTraits<T>;
+  using Vec = typename VT::Vec;
+  const long stride = (long)gridDim.x * blockDim.x;
+  if (VEC) {
+    const long nvec = total / VT::width;
+    for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < nvec;
+         i += stride) {
+      long base = i * VT::width;
+      int c0 = (int)(base % C);
+      Vec gv = reinterpret_cast<const Vec*>(dy)[i];
+      Vec xv = reinterpret_cast<const Vec*>(x)[i];
+      Vec yv;
+      if (RELU) yv = reinterpret_cast<const Vec*>(y)[i];
+      Vec ov;
+#pragma unroll
+      for (int j = 0; j < VT::width; ++j) {
+        int c = c0 + j;
+        float g = VT::load(gv, j);
+        if (RELU && VT::load(yv, j) <= 0.f) g = 0.f;
+        float xh = (VT::load(xv, j) - save_mean[c]) * save_invstd[c];
+        VT::store(ov, j, gi[c] * (g - k1[c] - xh * k2[c]));
+      }
+      reinterpret_cast<Vec*>(dx)[i] = ov;
+    }
+  } else {
+    for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < total;
+         i += stride) {
+      int c = (int)(i % C);
+      float g = VT::load1(dy + i);
+      if (RELU && VT::load1(y + i) <= 0.f) g = 0.f;
+      float xh = (VT::load1(x + i) - save_mean[c]) * save_invstd[c];
+      VT::store1(dx + i, gi[c] * (g - k1[c] - xh * k2[c]));
+    }
+  }
+}
+
+// ---- host-side helpers --------------------------------------------------
+
+struct BNShape {
+  long R;
+  int C;
+  int CT;
+  dim3 rgrid;   // reduce grid
+  int lds;      // reduce LDS bytes
+};
+
+static BNShape bn_shape(const at::Tensor& x) {
+  BNShape s;
+  s.C = x.size(1);
+  s.R = x.numel() / s.C;
+  s.CT = channel_tile(s.C);
+  int rows_grp = kBlock / s.CT;
+  long stripes = (s.R + rows_grp - 1) / rows_grp;
+  // enough blocks to fill 256 CUs x 8 XCDs without atomics blowup
+  int max_stripes = 2048 / std::max(1, s.C / s.CT);
+  if (stripes > max_stripes) stripes = max_stripes;
+  if (stripes < 1) stripes = 1;
+  s.rgrid = dim3((unsigned)stripes, (unsigned)(s.C / s.CT));
+  s.lds = 2 * s.CT * sizeof(float);
+  return s;
+}
+
+template <typename T>
+static const T* tp(const at::Tensor& t) {
+  return reinterpret_cast<const T*>(t.data_ptr());
+}
+template <typename T>
+static T* tpm(at::Tensor& t) {
+  return reinterpret_cast<T*>(t.data_ptr());
+}
+
+}  // namespace
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd_train_impl(
+    const at::Tensor& x_in, const at::Tensor& w, const at::Tensor& b,
+    at::Tensor& rmean, at::Tensor& rvar, double momentum, double eps,
+    bool relu) {
+  auto x = x_in.contiguous(at::MemoryFormat::ChannelsLast);
+  TORCH_CHECK(x.dim() == 4, "bn: 4D NHWC expected");
+  TORCH_CHECK(w.scalar_type() == at::kFloat, "bn: params must be fp32");
+  auto s = at::cuda::getCurrentCUDAStream();
+  auto sh = bn_shape(x);
+  auto fopts = x.options().dtype(at::kFloat);
+  auto sums = at::zeros({2 * sh.C}, fopts);
+  auto save_mean = at::empty({sh.C}, fopts);
+  auto save_invstd = at::empty({sh.C}, fopts);
+  auto scale = at::empty({sh.C}, fopts);
+  auto shift = at::empty({sh.C}, fopts);
+  auto y = at::empty_like(x);
+
+  const long total = x.numel();
+  const bool bf = x.scalar_type() == at::kBFloat16;
+  TORCH_CHECK(bf || x.scalar_type() == at::kFloat, "bn: bf16/fp32 only");
+
+#define DLX_BN_T(T)                                                          \
+  bn_reduce_fwd_kernel<T><<<sh.rgrid, kBlock, sh.lds, s>>>(                  \
+      tp<T>(x), sums.data_ptr<float>(), sh.R, sh.C, sh.CT);                  \
+  bn_finalize_fwd_kernel<<<(sh.C + 255) / 256, 256, 0, s>>>(                 \
+      sums.data_ptr<float>(), w.data_ptr<float>(), b.data_ptr<float>(),      \
+      rmean.data_ptr<float>(), rvar.data_ptr<float>(),                       \
+      save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),            \
+      scale.data_ptr<float>(), shift.data_ptr<float>(), sh.R, sh.C,          \
+      (float)momentum, (float)eps);                                          \
+  {                                                                          \
+    const bool vec = (sh.C % VecTraits<T>::width) == 0;                      \
+    int grid = grid_for(total / (vec ? VecTraits<T>::width : 1) + 1);        \
+    if (relu && vec)                                                         \
+      bn_norm_kernel<T, true, true><<<grid, kBlock, 0, s>>>(                 \
+          tp<T>(x), tpm<T>(y), scale.data_ptr<float>(),                      \
+          shift.data_ptr<float>(), total, sh.C);                             \
+    else if (relu)                                                           \
+      bn_norm_kernel<T, true, false><<<grid, kBlock, 0, s>>>(                \
+          tp<T>(x), tpm<T>(y), scale.data_ptr<float>(),                      \
+          shift.data_ptr<float>(), total, sh.C);                             \
+    else if (vec)                                                            \
+      bn_norm_kernel<T, false, true><<<grid, kBlock, 0, s>>>(                \
+          tp<T>(x), tpm<T>(y), scale.data_ptr<float>(),                      \
+          shift.data_ptr<float>(), total, sh.C);                             \
+    else                                                                     \
+      bn_norm_kernel<T, false, false><<<grid, kBlock, 0, s>>>(               \
+          tp<T>(x), tpm<T>(y), scale.data_ptr<float>(),                      \
+          shift.data_ptr<float>(), total, sh.C);                             \
+  }
+
+  if (bf) { DLX_BN_T(uint16_t) } else { DLX_BN_T(float) }
+#undef DLX_BN_T
+  return {y, save_mean, save_invstd};
+}
+
+at::Tensor bn_fwd_eval_impl(const at::Tensor& x_in, const at::Tensor& w,
+                            const at::Tensor& b, const at::Tensor& rmean,
+                            const at::Tensor& rvar, double eps, bool relu) {
+  auto x = x_in.contiguous(at::MemoryFormat::ChannelsLast);
+  auto s = at::cuda::getCurrentCUDAStream();
+  const int C = x.size(1);
+  const long total = x.numel();
+  auto fopts = x.options().dtype(at::kFloat);
+  auto scale = at::empty({C}, fopts);
+  auto shift = at::empty({C}, fopts);
+  bn_eval_coeffs_kernel<<<(C + 255) / 256, 256, 0, s>>>(
+      w.data_ptr<float>(), b.data_ptr<float>(), rmean.data_ptr<float>(),
+      rvar.data_ptr<float>(), scale.data_ptr<float>(), shift.data_ptr<float>(),
+      C, (float)eps);
+  auto y = at::empty_like(x);
+  const bool bf = x.scalar_type() == at::kBFloat16;
+
+#define DLX_BN_EVAL(T)                                                       \
+  {                                                                          \
+    const bool vec = (C % VecTraits<T>::width) == 0;                         \
+    int grid = grid_for(total / (vec ? VecTraits<T>::width : 1) + 1);        \
+    if (relu && vec)                                                         \
+      bn_norm_kernel<T, true, true><<<grid, kBlock, 0, s>>>(                 \
+          tp<T>(x), tpm<T>(y), scale.data_ptr<float>(),                      \
+          shift.data_ptr<float>(), total, C);                                \
+    else if (relu)                                                           \
+      bn_norm_kernel<T, true, false><<<grid, kBlock, 0, s>>>(                \
+          tp<T>(x), tpm<T>(y), scale.data_ptr<float>(),                      \
+          shift.data_ptr<float>(), total, C);                                \
+    else if (vec)                                                            \
+      bn_norm_kernel<T, false, true><<<grid, kBlock, 0, s>>>(                \
+          tp<T>(x), tpm<T>(y), scale.data_ptr<float>(),                      \
+          shift.data_ptr<float>(), total, C);                                \
+    else                                                                     \
+      bn_norm_kernel<T, false, false><<<grid, kBlock, 0, s>>>(               \
+          tp<T>(x), tpm<T>(y), scale.data_ptr<float>(),                      \
+          shift.data_ptr<float>(), total, C);                                \
+  }
+  if (bf) { DLX_BN_EVAL(uint16_t) } else { DLX_BN_EVAL(float) }
+#undef DLX_BN_EVAL
+  return y;
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_bwd_impl(
+    const at::Tensor& dy_in, const at::Tensor& x_in, const at::Tensor& w,
+    const at::Tensor& save_mean, const at::Tensor& save_invstd,
+    const at::Tensor& y_in, bool relu) {
+  auto x = x_in.contiguous(at::MemoryFormat::ChannelsLast);
+  auto dy = dy_in.contiguous(at::MemoryFormat::ChannelsLast);
+  auto y = y_in.contiguous(at::MemoryFormat::ChannelsLast);
+  auto s = at::cuda::getCurrentCUDAStream();
+  auto sh = bn_shape(x);
+  auto fopts = x.options().dtype(at::kFloat);
+  auto sums = at::zeros({2 * sh.C}, fopts);
+  auto dgamma = at::empty({sh.C}, fopts);
+  auto dbeta = at::empty({sh.C}, fopts);
+  auto k1 = at::empty({sh.C}, fopts);
+  auto k2 = at::empty({sh.C}, fopts);
+  auto gi = at::empty({sh.C}, fopts);
+  auto dx = at::empty_like(x);
+  const long total = x.numel();
+  const bool bf = x.scalar_type() == at::kBFloat16;
+
+#define DLX_BN_BWD(T)                                                        \
+  {                                                                          \
+    if (relu)                                                                \
+      bn_reduce_bwd_kernel<T, true><<<sh.rgrid, kBlock, sh.lds, s>>>(        \
+          tp<T>(dy), tp<T>(x), tp<T>(y), save_mean.data_ptr<float>(),        \
+          save_invstd.data_ptr<float>(), sums.data_ptr<float>(), sh.R,       \
+          sh.C, sh.CT);                                                      \
+    else                                                                     \
+      bn_reduce_bwd_kernel<T, false><<<sh.rgrid, kBlock, sh.lds, s>>>(       \
+          tp<T>(dy), tp<T>(x), tp<T>(y), save_mean.data_ptr<float>(),        \
+          save_invstd.data_ptr<float>(), sums.data_ptr<float>(), sh.R,       \
+          sh.C, sh.CT);                                                      \
+    bn_finalize_bwd_kernel<<<(sh.C + 255) / 256, 256, 0, s>>>(               \
+        sums.data_ptr<float>(), w.data_ptr<float>(),                         \
+        save_invstd.data_ptr<float>(), dgamma.data_ptr<float>(),             \
+        dbeta.data_ptr<float>(), k1.data_ptr<float>(),                       \
+        k2.data_ptr<float>(), gi.data_ptr<float>(), sh.R, sh.C);             \
+    const bool vec = (sh.C % VecTraits<T>::width) == 0;                      \
+    int grid = grid_for(total / (vec ? VecTraits<T>::width : 1) + 1);        \
+    if (relu && vec)                                                         \
+      bn_dx_kernel<T, true, true><<<grid, kBlock, 0, s>>>(                   \
+          tp<T>(dy), tp<T>(x), tp<T>(y), save_mean.data_ptr<float>(),        \
+          save_invstd.data_ptr<float>(), k1.data_ptr<float>(),               \
+          k2.data_ptr<float>(), gi.data_ptr<float>(), tpm<T>(dx), total,     \
+          sh.C);                                                             \
+    else if (relu)                                                           \
+      bn_dx_kernel<T, true, false><<<grid, kBlock, 0, s>>>(                  \
+          tp<T>(dy), tp<T>(x), tp<T>(y), save_mean.data_ptr<float>(),        \
+          save_invstd.data_ptr<float>(), k1.data_ptr<float>(),               \
+          k2.data_ptr<float>(), gi.data_ptr<float>(), tpm<T>(dx), total,     \
+          sh.C);                                                             \
+    else if (vec)                                                            \
+      bn_dx_kernel<T, false, true><<<grid, kBlock, 0, s>>>(                  \
+          tp<T>(dy), tp<T>(x), tp<T>(y), save_mean.data_ptr<float>(),        \
+          save_invstd.data_ptr<float>(), k1.data_ptr<float>(),               \
+          k2.data_ptr<float>(), gi.data_ptr<float>(), tpm<T>(dx), total,     \
+          sh.C);                                                             \
+    else                                                                     \
+      bn_dx_kernel<T, false, false><<<grid, kBlock, 0, s>>>(                 \
+          tp<T>(dy), tp<T>(x), tp<T>(y), save_mean.data_ptr<float>(),        \
+          save_invstd.data_ptr<float>(), k1.data_ptr<float>(),               \
+          k2.data_ptr<float>(), gi.data_ptr<float>(), tpm<T>(dx), total,     \
+          sh.C);                                                             \
+  }
+  if (bf) { DLX_BN_BWD(uint16_t) } else { DLX_BN_BWD(float) }
+#undef DLX_BN_BWD
+  return {dx, dgamma, dbeta};
+}
+
+}  // namespace dlx

@@ -1,0 +1,187 @@
+"""Dispatching functional layer: HIP-native on GPU, torch fallback on CPU.
+
+Each function is the seam the north star names: the torch path is the
+correctness reference (tests compare native vs a plain fp32 torch
+implementation of the same op), the HIP path is the production one.
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from . import native, native_available
+
+
+def _use_native(x: torch.Tensor) -> bool:
+    return x.is_cuda and native_available()
+
+
+def _to_nhwc(x: torch.Tensor) -> torch.Tensor:
+    return x.contiguous(memory_format=torch.channels_last)
+
+
+# ---------------------------------------------------------------------------
+# fused residual add + ReLU (tail of every ResNet block)
+# ---------------------------------------------------------------------------
+
+class _AddReLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+        out = native().add_relu_fwd(a, b)
+        ctx.save_for_backward(out)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad: torch.Tensor):
+        (out,) = ctx.saved_tensors
+        g = native().relu_bwd(grad, out)  # op contiguates to out's layout
+        return g, g
+
+
+def add_relu(a: torch.Tensor, b: torch.Tensor) -> torch.Tensor:
+    if _use_native(a) and a.dtype in (torch.bfloat16, torch.float16, torch.float32):
+        return _AddReLU.apply(a, b)
+    return F.relu(a + b)
+
+
+# ---------------------------------------------------------------------------
+# standalone ReLU (kept for completeness with the reference's op list)
+# ---------------------------------------------------------------------------
+
+class _ReLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x: torch.Tensor) -> torch.Tensor:
+        out = native().relu_fwd(x)
+        ctx.save_for_backward(out)
+        return out
+
+    @staticmethod
+    def backward(ctx, grad: torch.Tensor):
+        (out,) = ctx.saved_tensors
+        return native().relu_bwd(grad, out)
+
+
+def relu(x: torch.Tensor) -> torch.Tensor:
+    if _use_native(x):
+        return _ReLU.apply(x)
+    return F.relu(x)
+
+
+# ---------------------------------------------------------------------------
+# BatchNorm (+ fused ReLU) — NHWC-native, fp32 statistics
+# ---------------------------------------------------------------------------
+
+class _BNAct(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, running_mean, running_var,
+                momentum: float, eps: float, relu_: bool):
+        x = _to_nhwc(x)
+        y, save_mean, save_invstd = native().bn_fwd_train(
+            x, weight, bias, running_mean, running_var, momentum, eps, relu_)
+        ctx.save_for_backward(x, weight, save_mean, save_invstd, y)
+        ctx.relu = relu_
+        return y
+
+    @staticmethod
+    def backward(ctx, grad: torch.Tensor):
+        x, weight, save_mean, save_invstd, y = ctx.saved_tensors
+        dx, dweight, dbias = native().bn_bwd(
+            _to_nhwc(grad), x, weight, save_mean, save_invstd, y, ctx.relu)
+        return dx, dweight, dbias, None, None, None, None, None
+
+
+def batch_norm_act(x: torch.Tensor, bn: torch.nn.BatchNorm2d,
+                   relu: bool = True, training: bool = True) -> torch.Tensor:
+    """BatchNorm2d forward with optional fused ReLU.
+
+    Native path: one NHWC HIP kernel pair (Welford-style fp32 stats +
+    normalize+activate fused — replaces cuDNN BN + framework ReLU,
+    SURVEY.md §3.2 rows cuDNN-BatchNorm / framework-elementwise).
+    """
+    if training and _use_native(x):
+        return _BNAct.apply(x, bn.weight, bn.bias, bn.running_mean,
+                            bn.running_var, bn.momentum, bn.eps, relu)
+    if not training and _use_native(x):
+        return native().bn_fwd_eval(_to_nhwc(x), bn.weight, bn.bias,
+                                    bn.running_mean, bn.running_var,
+                                    bn.eps, relu)
+    out = F.batch_norm(x, bn.running_mean, bn.running_var, bn.weight,
+                       bn.bias, training, bn.momentum, bn.eps)
+    return F.relu(out) if relu else out
+
+
+# ---------------------------------------------------------------------------
+# fused softmax + cross-entropy (the classifier head)
+# ---------------------------------------------------------------------------
+
+class _SoftmaxCE(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, logits: torch.Tensor, target: torch.Tensor):
+        loss, probs = native().softmax_ce_fwd(logits.contiguous(), target)
+        ctx.save_for_backward(probs, target)
+        ctx.out_bf16 = logits.dtype == torch.bfloat16
+        return loss
+
+    @staticmethod
+    def backward(ctx, grad: torch.Tensor):
+        probs, target = ctx.saved_tensors
+        return native().softmax_ce_bwd(probs, target, grad, ctx.out_bf16), None
+
+
+def softmax_cross_entropy(logits: torch.Tensor,
+                          target: torch.Tensor) -> torch.Tensor:
+    """Mean cross-entropy over the batch (fused softmax+CE on HIP)."""
+    if _use_native(logits) and logits.dim() == 2:
+        return _SoftmaxCE.apply(logits, target)
+    return F.cross_entropy(logits.float(), target)
+
+
+# ---------------------------------------------------------------------------
+# layout pack (NHWC <-> NCHW) — explicit north-star op
+# ---------------------------------------------------------------------------
+
+def nchw_to_nhwc(x: torch.Tensor) -> torch.Tensor:
+    if _use_native(x) and x.dim() == 4:
+        return native().pack_nchw_to_nhwc(x)
+    return x.contiguous(memory_format=torch.channels_last)
+
+
+def nhwc_to_nchw(x: torch.Tensor) -> torch.Tensor:
+    if _use_native(x) and x.dim() == 4:
+        return native().pack_nhwc_to_nchw(x)
+    return x.contiguous(memory_format=torch.contiguous_format)
+
+
+# ---------------------------------------------------------------------------
+# GEMM / Linear (MFMA kernel; used by the FC head and the BERT path)
+# ---------------------------------------------------------------------------
+
+class _LinearMFMA(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        ctx.save_for_backward(x, weight)
+        ctx.has_bias = bias is not None
+        y = native().gemm_nt(x, weight)  # [M,K] @ [N,K]^T -> [M,N]
+        if bias is not None:
+            y = y + bias
+        return y
+
+    @staticmethod
+    def backward(ctx, grad):
+        x, weight = ctx.saved_tensors
+        grad = grad.contiguous()
+        dx = native().gemm_nn(grad, weight)        # [M,N] @ [N,K] -> [M,K]
+        dw = native().gemm_tn(grad, x)             # [M,N]^T @ [M,K] -> [N,K]
+        db = grad.sum(0) if ctx.has_bias else None
+        return dx, dw, db
+
+
+def linear(x: torch.Tensor, weight: torch.Tensor,
+           bias: Optional[torch.Tensor] = None) -> torch.Tensor:
+    if (_use_native(x) and x.dtype == torch.bfloat16 and x.dim() == 2
+            and x.shape[1] % 16 == 0 and weight.shape[0] % 16 == 0):
+        return _LinearMFMA.apply(x, weight, bias)
+    return F.linear(x, weight, bias)
