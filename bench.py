@@ -66,8 +66,8 @@ def main() -> int:
     model = model.to(ctx.device)
     if cuda:
         model = model.to(memory_format=torch.channels_last)
-    if dtype != torch.float32:
-        model = model.to(dtype)
+    from deeplearning_cfn_amd.utils import to_compute_dtype
+    model = to_compute_dtype(model, dtype)
     engine = DataParallel(model, mode="dist_device_sync")
     opt = FlatSGD(engine, lr=0.1, momentum=0.9, weight_decay=1e-4)
 

@@ -125,17 +125,19 @@ class DataParallel:
         Buckets never mix dtypes or devices.
         """
         plan = BucketPlan()
-        cur: List[torch.nn.Parameter] = []
-        cur_bytes = 0
-        key = None
+        # one OPEN bucket per (dtype, device): mixed-precision models (bf16
+        # weights + fp32 norm params) interleave dtypes in parameter order,
+        # and flushing on every transition would fragment the collectives.
+        open_buckets: Dict[tuple, List[torch.nn.Parameter]] = {}
+        open_bytes: Dict[tuple, int] = {}
 
-        def flush():
-            nonlocal cur, cur_bytes
+        def flush(k):
+            cur = open_buckets.pop(k, [])
+            open_bytes.pop(k, None)
             if not cur:
                 return
             n = sum(p.numel() for p in cur)
-            flat = torch.zeros(
-                n, dtype=cur[0].dtype, device=cur[0].device)
+            flat = torch.zeros(n, dtype=cur[0].dtype, device=cur[0].device)
             views, off = [], 0
             for p in cur:
                 v = flat[off:off + p.numel()].view_as(p)
@@ -147,16 +149,15 @@ class DataParallel:
             for p, v in zip(b.params, b.views):
                 p.grad = v
             plan.buckets.append(b)
-            cur, cur_bytes = [], 0
 
         for p in reversed(params):
             k = (p.dtype, p.device)
-            if key is not None and (k != key or cur_bytes >= self.bucket_bytes):
-                flush()
-            key = k
-            cur.append(p)
-            cur_bytes += p.numel() * p.element_size()
-        flush()
+            open_buckets.setdefault(k, []).append(p)
+            open_bytes[k] = open_bytes.get(k, 0) + p.numel() * p.element_size()
+            if open_bytes[k] >= self.bucket_bytes:
+                flush(k)
+        for k in list(open_buckets):
+            flush(k)
         return plan
 
     def _install_hooks(self) -> None:

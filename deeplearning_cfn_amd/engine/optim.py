@@ -28,6 +28,9 @@ class _FlatOptimBase:
         # one contiguous param mirror per bucket so the fused kernel walks
         # a single pointer; params are re-viewed into it (zero-copy update).
         self.flat_params: List[torch.Tensor] = []
+        # fp32 master weights for reduced-precision params (bf16 training):
+        # the fused kernels update the master and write the bf16 copy.
+        self.masters: List[Optional[torch.Tensor]] = []
         for b in self.buckets:
             flat_p = torch.empty_like(b.flat)
             off = 0
@@ -38,6 +41,8 @@ class _FlatOptimBase:
                     p.data = flat_p[off:off + p.numel()].view_as(p)
                     off += p.numel()
             self.flat_params.append(flat_p)
+            self.masters.append(flat_p.float()
+                                if flat_p.dtype != torch.float32 else None)
 
     def zero_grad(self) -> None:
         self.engine.zero_grad()
@@ -62,31 +67,44 @@ class FlatSGD(_FlatOptimBase):
 
     @torch.no_grad()
     def step(self) -> None:
-        for fp, b, mbuf in zip(self.flat_params, self.buckets,
-                               self.momentum_bufs):
+        for fp, b, mbuf, master in zip(self.flat_params, self.buckets,
+                                       self.momentum_bufs, self.masters):
             g = b.flat
             if fp.is_cuda and native_available():
-                native().sgd_momentum_flat(
-                    fp, g, mbuf, self.lr, self.momentum,
-                    self.weight_decay, self.nesterov)
+                if master is not None:
+                    native().sgd_momentum_flat_master(
+                        fp, g, mbuf, master, self.lr, self.momentum,
+                        self.weight_decay, self.nesterov)
+                else:
+                    native().sgd_momentum_flat(
+                        fp, g, mbuf, self.lr, self.momentum,
+                        self.weight_decay, self.nesterov)
                 continue
             # torch fallback — numerically the reference update:
             # m = mu*m + (g + wd*p); p -= lr * (m or g+mu*m)
+            pv = master if master is not None else fp
             g32 = g.float()
             if self.weight_decay:
-                g32 = g32.add(fp.float(), alpha=self.weight_decay)
+                g32 = g32.add(pv.float(), alpha=self.weight_decay)
             mbuf.mul_(self.momentum).add_(g32)
             upd = g32.add(mbuf, alpha=self.momentum) if self.nesterov else mbuf
-            fp.copy_((fp.float() - self.lr * upd).to(fp.dtype))
+            if master is not None:
+                master.sub_(self.lr * upd)
+                fp.copy_(master.to(fp.dtype))
+            else:
+                fp.copy_((fp.float() - self.lr * upd).to(fp.dtype))
 
     def state_dict(self) -> dict:
-        return {"momentum_bufs": self.momentum_bufs,
+        return {"momentum_bufs": self.momentum_bufs, "masters": self.masters,
                 "lr": self.lr, "momentum": self.momentum,
                 "weight_decay": self.weight_decay, "nesterov": self.nesterov}
 
     def load_state_dict(self, sd: dict) -> None:
         for dst, src in zip(self.momentum_bufs, sd["momentum_bufs"]):
             dst.copy_(src.to(dst.device))
+        for dst, src in zip(self.masters, sd.get("masters", [])):
+            if dst is not None and src is not None:
+                dst.copy_(src.to(dst.device))
         self.lr = sd["lr"]
         self.momentum = sd["momentum"]
         self.weight_decay = sd["weight_decay"]
@@ -113,18 +131,26 @@ class FlatAdam(_FlatOptimBase):
         b1, b2 = self.betas
         bc1 = 1 - b1 ** self.t
         bc2 = 1 - b2 ** self.t
-        for fp, b, m, v in zip(self.flat_params, self.buckets, self.m, self.v):
+        for fp, b, m, v, master in zip(self.flat_params, self.buckets,
+                                       self.m, self.v, self.masters):
             g = b.flat
             if fp.is_cuda and native_available():
-                native().adamw_flat(fp, g, m, v, self.lr, b1, b2, self.eps,
-                                    self.weight_decay, bc1, bc2)
+                if master is not None:
+                    native().adamw_flat_master(fp, g, m, v, master, self.lr,
+                                               b1, b2, self.eps,
+                                               self.weight_decay, bc1, bc2)
+                else:
+                    native().adamw_flat(fp, g, m, v, self.lr, b1, b2,
+                                        self.eps, self.weight_decay, bc1, bc2)
                 continue
             g32 = g.float()
             m.mul_(b1).add_(g32, alpha=1 - b1)
             v.mul_(b2).addcmul_(g32, g32, value=1 - b2)
-            p32 = fp.float()
+            p32 = master if master is not None else fp.float()
             p32 = p32 - self.lr * self.weight_decay * p32
             p32 = p32 - self.lr * (m / bc1) / ((v / bc2).sqrt() + self.eps)
+            if master is not None:
+                master.copy_(p32)
             fp.copy_(p32.to(fp.dtype))
 
     def state_dict(self) -> dict:

@@ -97,8 +97,8 @@ def train(args: argparse.Namespace) -> dict:
     model = model.to(ctx.device)
     if not is_bert:
         model = model.to(memory_format=torch.channels_last)
-    if dtype != torch.float32:
-        model = model.to(dtype)
+    from .utils import to_compute_dtype
+    model = to_compute_dtype(model, dtype)
 
     bucket_bytes = int(args.bucket_mb * (1 << 20)) if args.bucket_mb else None
     engine = DataParallel(model, bucket_bytes=bucket_bytes, mode=mode)

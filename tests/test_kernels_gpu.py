@@ -1,0 +1,182 @@
+"""HIP kernel numerics vs plain fp32 torch references (GPU-only).
+
+Every native op is compared against the torch fp32 implementation of the
+same op on random tensors.  Inputs are random
+(never symmetric/zero — guide §5.4 rules 16/25).
+"""
+
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+def _native():
+    from deeplearning_cfn_amd.ops import native
+    return native()
+
+
+def _rand(shape, dtype=torch.bfloat16, cl=False, seed=0):
+    g = torch.Generator(device="cuda").manual_seed(seed)
+    x = torch.randn(*shape, generator=g, device="cuda", dtype=torch.float32)
+    x = x.to(dtype)
+    if cl:
+        x = x.contiguous(memory_format=torch.channels_last)
+    return x
+
+
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
+def test_relu_fwd_bwd(dtype):
+    n = _native()
+    x = _rand((33, 129), dtype)  # odd sizes exercise the scalar tail
+    y = n.relu_fwd(x)
+    torch.testing.assert_close(y.float(), torch.relu(x.float()))
+    g = _rand((33, 129), dtype, seed=1)
+    dx = n.relu_bwd(g, y)
+    torch.testing.assert_close(dx.float(),
+                               g.float() * (y.float() > 0))
+
+
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
+def test_add_relu(dtype):
+    n = _native()
+    a = _rand((8, 64, 14, 14), dtype, cl=True)
+    b = _rand((8, 64, 14, 14), dtype, cl=True, seed=2)
+    y = n.add_relu_fwd(a, b)
+    ref = torch.relu(a.float() + b.float())
+    tol = 2e-2 if dtype == torch.bfloat16 else 1e-6
+    torch.testing.assert_close(y.float(), ref, rtol=tol, atol=tol)
+
+
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
+@pytest.mark.parametrize("relu", [False, True])
+@pytest.mark.parametrize("C", [16, 64, 100, 256])
+def test_bn_fwd_train_parity(dtype, relu, C):
+    n = _native()
+    x = _rand((4, C, 7, 9), dtype, cl=True, seed=C)
+    w = torch.randn(C, device="cuda") * 0.5 + 1.0
+    b = torch.randn(C, device="cuda") * 0.1
+    rmean = torch.zeros(C, device="cuda")
+    rvar = torch.ones(C, device="cuda")
+    rmean_ref, rvar_ref = rmean.clone(), rvar.clone()
+
+    y, mean, invstd = n.bn_fwd_train(x, w, b, rmean, rvar, 0.1, 1e-5, relu)
+    ref = torch.nn.functional.batch_norm(
+        x.float(), rmean_ref, rvar_ref, w, b, True, 0.1, 1e-5)
+    if relu:
+        ref = torch.relu(ref)
+    tol = 3e-2 if dtype == torch.bfloat16 else 1e-4
+    torch.testing.assert_close(y.float(), ref, rtol=tol, atol=tol)
+    torch.testing.assert_close(rmean, rmean_ref, rtol=1e-3, atol=1e-4)
+    torch.testing.assert_close(rvar, rvar_ref, rtol=1e-3, atol=1e-4)
+
+
+@pytest.mark.parametrize("relu", [False, True])
+def test_bn_bwd_parity(relu):
+    """Native BN bwd vs autograd through fp32 F.batch_norm."""
+    n = _native()
+    C = 32
+    xf = _rand((4, C, 6, 6), torch.float32, cl=True, seed=5)
+    w = (torch.randn(C, device="cuda") * 0.5 + 1.0).requires_grad_()
+    b = (torch.randn(C, device="cuda") * 0.1).requires_grad_()
+    x = xf.clone().requires_grad_()
+    rm, rv = torch.zeros(C, device="cuda"), torch.ones(C, device="cuda")
+
+    ref = torch.nn.functional.batch_norm(x, rm.clone(), rv.clone(), w, b,
+                                         True, 0.1, 1e-5)
+    if relu:
+        ref = torch.relu(ref)
+    dy = _rand((4, C, 6, 6), torch.float32, cl=True, seed=6)
+    ref.backward(dy)
+
+    y, mean, invstd = n.bn_fwd_train(xf, w.detach(), b.detach(), rm, rv,
+                                     0.1, 1e-5, relu)
+    dx, dw, db = n.bn_bwd(dy, xf, w.detach(), mean, invstd, y, relu)
+    torch.testing.assert_close(dx.float(), x.grad, rtol=1e-3, atol=1e-4)
+    torch.testing.assert_close(dw, w.grad, rtol=1e-3, atol=1e-3)
+    torch.testing.assert_close(db, b.grad, rtol=1e-3, atol=1e-3)
+
+
+def test_bn_eval_parity():
+    n = _native()
+    C = 64
+    x = _rand((2, C, 8, 8), torch.bfloat16, cl=True, seed=9)
+    w = torch.randn(C, device="cuda") * 0.5 + 1.0
+    b = torch.randn(C, device="cuda") * 0.1
+    rm = torch.randn(C, device="cuda") * 0.2
+    rv = torch.rand(C, device="cuda") + 0.5
+    y = n.bn_fwd_eval(x, w, b, rm, rv, 1e-5, True)
+    ref = torch.relu(torch.nn.functional.batch_norm(
+        x.float(), rm, rv, w, b, False, 0.1, 1e-5))
+    torch.testing.assert_close(y.float(), ref, rtol=3e-2, atol=3e-2)
+
+
+@pytest.mark.parametrize("dtype", [torch.bfloat16, torch.float32])
+def test_softmax_ce(dtype):
+    n = _native()
+    N, C = 64, 1000
+    logits = _rand((N, C), dtype, seed=3)
+    target = torch.randint(0, C, (N,), device="cuda")
+    loss, probs = n.softmax_ce_fwd(logits, target)
+    ref = torch.nn.functional.cross_entropy(logits.float(), target)
+    torch.testing.assert_close(loss, ref, rtol=1e-3, atol=1e-4)
+    # backward
+    g = torch.tensor(2.0, device="cuda")
+    dl = n.softmax_ce_bwd(probs, target, g, dtype == torch.bfloat16)
+    lref = logits.float().detach().requires_grad_()
+    torch.nn.functional.cross_entropy(lref, target).backward(
+        torch.tensor(2.0, device="cuda"))
+    tol = 2e-2 if dtype == torch.bfloat16 else 1e-5
+    torch.testing.assert_close(dl.float(), lref.grad, rtol=tol, atol=tol)
+    assert dl.dtype == dtype
+
+
+def test_sgd_flat_fp32_matches_torch():
+    n = _native()
+    N = 10007  # odd: exercises tail
+    p = torch.randn(N, device="cuda")
+    g = torch.randn(N, device="cuda")
+    m = torch.randn(N, device="cuda").abs()
+    p_ref = p.clone().requires_grad_()
+    opt = torch.optim.SGD([p_ref], lr=0.05, momentum=0.9, weight_decay=1e-3)
+    opt.state[p_ref]["momentum_buffer"] = m.clone()
+    p_ref.grad = g.clone()
+    opt.step()
+    n.sgd_momentum_flat(p, g, m, 0.05, 0.9, 1e-3, False)
+    torch.testing.assert_close(p, p_ref.detach(), rtol=1e-6, atol=1e-6)
+
+
+def test_sgd_flat_bf16_master():
+    n = _native()
+    N = 4096
+    master = torch.randn(N, device="cuda")
+    p = master.bfloat16()
+    g = (torch.randn(N, device="cuda") * 0.1).bfloat16()
+    m = torch.zeros(N, device="cuda")
+    master_ref = master.clone()
+    m_ref = torch.zeros(N, device="cuda")
+    for _ in range(3):
+        n.sgd_momentum_flat_master(p, g, m, master, 0.1, 0.9, 1e-4, False)
+        gr = g.float() + 1e-4 * master_ref
+        m_ref.mul_(0.9).add_(gr)
+        master_ref -= 0.1 * m_ref
+    torch.testing.assert_close(master, master_ref, rtol=1e-5, atol=1e-6)
+    torch.testing.assert_close(p.float(), master.bfloat16().float())
+
+
+def test_adamw_flat_matches_torch():
+    n = _native()
+    N = 8191
+    p = torch.randn(N, device="cuda")
+    g = torch.randn(N, device="cuda")
+    m = torch.zeros(N, device="cuda")
+    v = torch.zeros(N, device="cuda")
+    p_ref = p.clone().requires_grad_()
+    opt = torch.optim.AdamW([p_ref], lr=1e-3, betas=(0.9, 0.999), eps=1e-8,
+                            weight_decay=0.01)
+    for t in range(1, 4):
+        p_ref.grad = g.clone()
+        opt.step()
+        bc1, bc2 = 1 - 0.9 ** t, 1 - 0.999 ** t
+        n.adamw_flat(p, g, m, v, 1e-3, 0.9, 0.999, 1e-8, 0.01, bc1, bc2)
+    torch.testing.assert_close(p, p_ref.detach(), rtol=1e-5, atol=1e-6)
