@@ -33,6 +33,10 @@ at::Tensor softmax_ce_bwd_impl(const at::Tensor& probs,
                                const at::Tensor& grad,
                                at::ScalarType out_dtype);
 
+// gemm.hip
+at::Tensor gemm_nt_impl(const at::Tensor& A, const at::Tensor& B,
+                        bool out_bf16);
+
 // batchnorm.hip
 std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd_train_impl(
     const at::Tensor& x, const at::Tensor& w, const at::Tensor& b,
@@ -157,6 +161,21 @@ at::Tensor softmax_ce_bwd(const at::Tensor& probs, const at::Tensor& target,
                              out_bf16 ? at::kBFloat16 : at::kFloat);
 }
 
+// x[M,K] @ W[N,K]^T -> [M,N]  (the natural MFMA form; both K-contiguous)
+at::Tensor gemm_nt(const at::Tensor& a, const at::Tensor& b) {
+  return gemm_nt_impl(a, b, /*out_bf16=*/true);
+}
+
+// g[M,N] @ W[N,K] -> [M,K]  (dgrad; W transposed to K-contiguous form)
+at::Tensor gemm_nn(const at::Tensor& g, const at::Tensor& w) {
+  return gemm_nt_impl(g, w.t().contiguous(), true);
+}
+
+// g[M,N]^T @ x[M,K] -> [N,K]  (wgrad)
+at::Tensor gemm_tn(const at::Tensor& g, const at::Tensor& x) {
+  return gemm_nt_impl(g.t().contiguous(), x.t().contiguous(), true);
+}
+
 }  // namespace
 }  // namespace dlx
 
@@ -173,6 +192,9 @@ TORCH_LIBRARY(dlx_hip, m) {
   m.def("bn_fwd_train(Tensor x, Tensor w, Tensor b, Tensor(a!) rmean, Tensor(b!) rvar, float momentum, float eps, bool relu) -> (Tensor, Tensor, Tensor)");
   m.def("bn_fwd_eval(Tensor x, Tensor w, Tensor b, Tensor rmean, Tensor rvar, float eps, bool relu) -> Tensor");
   m.def("bn_bwd(Tensor dy, Tensor x, Tensor w, Tensor save_mean, Tensor save_invstd, Tensor y, bool relu) -> (Tensor, Tensor, Tensor)");
+  m.def("gemm_nt(Tensor a, Tensor b) -> Tensor");
+  m.def("gemm_nn(Tensor g, Tensor w) -> Tensor");
+  m.def("gemm_tn(Tensor g, Tensor x) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(dlx_hip, CUDA, m) {
@@ -188,4 +210,7 @@ TORCH_LIBRARY_IMPL(dlx_hip, CUDA, m) {
   m.impl("bn_fwd_train", dlx::bn_fwd_train_impl);
   m.impl("bn_fwd_eval", dlx::bn_fwd_eval_impl);
   m.impl("bn_bwd", dlx::bn_bwd_impl);
+  m.impl("gemm_nt", dlx::gemm_nt);
+  m.impl("gemm_nn", dlx::gemm_nn);
+  m.impl("gemm_tn", dlx::gemm_tn);
 }
