@@ -33,6 +33,18 @@ def conv1x1(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
     return nn.Conv2d(cin, cout, 1, stride=stride, bias=False)
 
 
+class StemMaxPool(nn.Module):
+    """3x3/s2/p1 maxpool (native HIP NHWC kernel with argmax backward)."""
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return F_amd.maxpool3x3s2(x)
+
+
+class GlobalAvgPool(nn.Module):
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return F_amd.global_avg_pool(x)
+
+
 class BNReLU(nn.Module):
     """BatchNorm2d + ReLU, fused on the native path (ops/bn_relu kernel)."""
 
@@ -97,7 +109,7 @@ class ResNet(nn.Module):
         if stem == "imagenet":
             self.conv1 = nn.Conv2d(3, self.inplanes, 7, stride=2, padding=3,
                                    bias=False)
-            self.maxpool = nn.MaxPool2d(3, stride=2, padding=1)
+            self.maxpool = StemMaxPool()
         else:  # cifar: single 3x3 stem, no maxpool
             self.conv1 = conv3x3(3, self.inplanes)
             self.maxpool = nn.Identity()
@@ -109,7 +121,7 @@ class ResNet(nn.Module):
             stages.append(self._make_stage(block, planes * (2 ** i), n,
                                            stride=1 if i == 0 else 2))
         self.stages = nn.Sequential(*stages)
-        self.avgpool = nn.AdaptiveAvgPool2d(1)
+        self.avgpool = GlobalAvgPool()
         self.fc = nn.Linear(self.inplanes, num_classes)
 
         for m in self.modules():

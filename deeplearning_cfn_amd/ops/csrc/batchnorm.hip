@@ -27,6 +27,48 @@ static int channel_tile(int C) {
 
 // ---- reduction: sum(x), sum(x^2) per channel ---------------------------
 
+// Vectorized variant (C % VW == 0): each thread owns a VW-channel chunk
+// (16-B loads, guide G13); CT counts CHUNKS here.
+template <typename T>
+__global__ void bn_reduce_fwd_vec_kernel(const T* __restrict__ x,
+                                         float* __restrict__ sums,  // [2C]
+                                         long R, int C, int CT) {
+  using VT = VecTraits<T>;
+  using Vec = typename VT::Vec;
+  constexpr int VW = VT::width;
+  extern __shared__ float lds[];          // [2*CT*VW]
+  for (int i = threadIdx.x; i < 2 * CT * VW; i += blockDim.x) lds[i] = 0.f;
+  __syncthreads();
+
+  const int chunk = threadIdx.x % CT;
+  const int c0 = blockIdx.y * CT * VW + chunk * VW;
+  const int sub = threadIdx.x / CT;
+  const int rows_grp = blockDim.x / CT;
+  float s[VW] = {}, sq[VW] = {};
+  for (long r = (long)blockIdx.x * rows_grp + sub; r < R;
+       r += (long)gridDim.x * rows_grp) {
+    Vec v = *reinterpret_cast<const Vec*>(x + r * C + c0);
+#pragma unroll
+    for (int j = 0; j < VW; ++j) {
+      float f = VT::load(v, j);
+      s[j] += f;
+      sq[j] += f * f;
+    }
+  }
+  float* lsum = lds + chunk * VW;
+  float* lsq = lds + CT * VW + chunk * VW;
+#pragma unroll
+  for (int j = 0; j < VW; ++j) {
+    atomicAdd(&lsum[j], s[j]);
+    atomicAdd(&lsq[j], sq[j]);
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < CT * VW; i += blockDim.x) {
+    atomicAdd(&sums[blockIdx.y * CT * VW + i], lds[i]);
+    atomicAdd(&sums[C + blockIdx.y * CT * VW + i], lds[CT * VW + i]);
+  }
+}
+
 template <typename T>
 __global__ void bn_reduce_fwd_kernel(const T* __restrict__ x,
                                      float* __restrict__ sums,  // [2C]
@@ -138,6 +180,62 @@ __global__ void bn_norm_kernel(const T* __restrict__ x, T* __restrict__ y,
 // ---- backward reduce: sum(dy), sum(dy * xhat) per channel --------------
 
 template <typename T, bool RELU>
+__global__ void bn_reduce_bwd_vec_kernel(const T* __restrict__ dy,
+                                         const T* __restrict__ x,
+                                         const T* __restrict__ y,
+                                         const float* __restrict__ save_mean,
+                                         const float* __restrict__ save_invstd,
+                                         float* __restrict__ sums,  // [2C]
+                                         long R, int C, int CT) {
+  using VT = VecTraits<T>;
+  using Vec = typename VT::Vec;
+  constexpr int VW = VT::width;
+  extern __shared__ float lds[];
+  for (int i = threadIdx.x; i < 2 * CT * VW; i += blockDim.x) lds[i] = 0.f;
+  __syncthreads();
+
+  const int chunk = threadIdx.x % CT;
+  const int c0 = blockIdx.y * CT * VW + chunk * VW;
+  const int sub = threadIdx.x / CT;
+  const int rows_grp = blockDim.x / CT;
+  float mean[VW], invstd[VW];
+#pragma unroll
+  for (int j = 0; j < VW; ++j) {
+    mean[j] = save_mean[c0 + j];
+    invstd[j] = save_invstd[c0 + j];
+  }
+  float a[VW] = {}, bx[VW] = {};
+  for (long r = (long)blockIdx.x * rows_grp + sub; r < R;
+       r += (long)gridDim.x * rows_grp) {
+    const long i = r * C + c0;
+    Vec gv = *reinterpret_cast<const Vec*>(dy + i);
+    Vec xv = *reinterpret_cast<const Vec*>(x + i);
+    Vec yv;
+    if (RELU) yv = *reinterpret_cast<const Vec*>(y + i);
+#pragma unroll
+    for (int j = 0; j < VW; ++j) {
+      float g = VT::load(gv, j);
+      if (RELU && VT::load(yv, j) <= 0.f) g = 0.f;
+      float xh = (VT::load(xv, j) - mean[j]) * invstd[j];
+      a[j] += g;
+      bx[j] += g * xh;
+    }
+  }
+  float* s_dy = lds + chunk * VW;
+  float* s_dyx = lds + CT * VW + chunk * VW;
+#pragma unroll
+  for (int j = 0; j < VW; ++j) {
+    atomicAdd(&s_dy[j], a[j]);
+    atomicAdd(&s_dyx[j], bx[j]);
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < CT * VW; i += blockDim.x) {
+    atomicAdd(&sums[blockIdx.y * CT * VW + i], lds[i]);
+    atomicAdd(&sums[C + blockIdx.y * CT * VW + i], lds[CT * VW + i]);
+  }
+}
+
+template <typename T, bool RELU>
 __global__ void bn_reduce_bwd_kernel(const T* __restrict__ dy,
                                      const T* __restrict__ x,
                                      const T* __restrict__ y,
@@ -245,23 +343,26 @@ struct BNShape {
   long R;
   int C;
   int CT;
+  bool vec;     // vectorized reduce path (CT counts VW-chunks)
   dim3 rgrid;   // reduce grid
   int lds;      // reduce LDS bytes
 };
 
-static BNShape bn_shape(const at::Tensor& x) {
+static BNShape bn_shape(const at::Tensor& x, int vw) {
   BNShape s;
   s.C = x.size(1);
   s.R = x.numel() / s.C;
-  s.CT = channel_tile(s.C);
+  s.vec = (s.C % vw) == 0;
+  const int units = s.vec ? s.C / vw : s.C;   // per-thread ownership units
+  s.CT = channel_tile(units);
   int rows_grp = kBlock / s.CT;
   long stripes = (s.R + rows_grp - 1) / rows_grp;
   // enough blocks to fill 256 CUs x 8 XCDs without atomics blowup
-  int max_stripes = 2048 / std::max(1, s.C / s.CT);
+  int max_stripes = 2048 / std::max(1, units / s.CT);
   if (stripes > max_stripes) stripes = max_stripes;
   if (stripes < 1) stripes = 1;
-  s.rgrid = dim3((unsigned)stripes, (unsigned)(s.C / s.CT));
-  s.lds = 2 * s.CT * sizeof(float);
+  s.rgrid = dim3((unsigned)stripes, (unsigned)(units / s.CT));
+  s.lds = 2 * s.CT * (s.vec ? vw : 1) * sizeof(float);
   return s;
 }
 
@@ -284,7 +385,9 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd_train_impl(
   TORCH_CHECK(x.dim() == 4, "bn: 4D NHWC expected");
   TORCH_CHECK(w.scalar_type() == at::kFloat, "bn: params must be fp32");
   auto s = at::cuda::getCurrentCUDAStream();
-  auto sh = bn_shape(x);
+  const bool bf = x.scalar_type() == at::kBFloat16;
+  TORCH_CHECK(bf || x.scalar_type() == at::kFloat, "bn: bf16/fp32 only");
+  auto sh = bn_shape(x, bf ? 8 : 4);
   auto fopts = x.options().dtype(at::kFloat);
   auto sums = at::zeros({2 * sh.C}, fopts);
   auto save_mean = at::empty({sh.C}, fopts);
@@ -294,12 +397,14 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd_train_impl(
   auto y = at::empty_like(x);
 
   const long total = x.numel();
-  const bool bf = x.scalar_type() == at::kBFloat16;
-  TORCH_CHECK(bf || x.scalar_type() == at::kFloat, "bn: bf16/fp32 only");
 
 #define DLX_BN_T(T)                                                          \
-  bn_reduce_fwd_kernel<T><<<sh.rgrid, kBlock, sh.lds, s>>>(                  \
-      tp<T>(x), sums.data_ptr<float>(), sh.R, sh.C, sh.CT);                  \
+  if (sh.vec)                                                                \
+    bn_reduce_fwd_vec_kernel<T><<<sh.rgrid, kBlock, sh.lds, s>>>(            \
+        tp<T>(x), sums.data_ptr<float>(), sh.R, sh.C, sh.CT);                \
+  else                                                                       \
+    bn_reduce_fwd_kernel<T><<<sh.rgrid, kBlock, sh.lds, s>>>(                \
+        tp<T>(x), sums.data_ptr<float>(), sh.R, sh.C, sh.CT);                \
   bn_finalize_fwd_kernel<<<(sh.C + 255) / 256, 256, 0, s>>>(                 \
       sums.data_ptr<float>(), w.data_ptr<float>(), b.data_ptr<float>(),      \
       rmean.data_ptr<float>(), rvar.data_ptr<float>(),                       \
@@ -383,7 +488,8 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_bwd_impl(
   auto dy = dy_in.contiguous(at::MemoryFormat::ChannelsLast);
   auto y = y_in.contiguous(at::MemoryFormat::ChannelsLast);
   auto s = at::cuda::getCurrentCUDAStream();
-  auto sh = bn_shape(x);
+  const bool bf = x.scalar_type() == at::kBFloat16;
+  auto sh = bn_shape(x, bf ? 8 : 4);
   auto fopts = x.options().dtype(at::kFloat);
   auto sums = at::zeros({2 * sh.C}, fopts);
   auto dgamma = at::empty({sh.C}, fopts);
@@ -393,11 +499,20 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_bwd_impl(
   auto gi = at::empty({sh.C}, fopts);
   auto dx = at::empty_like(x);
   const long total = x.numel();
-  const bool bf = x.scalar_type() == at::kBFloat16;
 
 #define DLX_BN_BWD(T)                                                        \
   {                                                                          \
-    if (relu)                                                                \
+    if (sh.vec && relu)                                                      \
+      bn_reduce_bwd_vec_kernel<T, true><<<sh.rgrid, kBlock, sh.lds, s>>>(    \
+          tp<T>(dy), tp<T>(x), tp<T>(y), save_mean.data_ptr<float>(),        \
+          save_invstd.data_ptr<float>(), sums.data_ptr<float>(), sh.R,       \
+          sh.C, sh.CT);                                                      \
+    else if (sh.vec)                                                         \
+      bn_reduce_bwd_vec_kernel<T, false><<<sh.rgrid, kBlock, sh.lds, s>>>(   \
+          tp<T>(dy), tp<T>(x), tp<T>(y), save_mean.data_ptr<float>(),        \
+          save_invstd.data_ptr<float>(), sums.data_ptr<float>(), sh.R,       \
+          sh.C, sh.CT);                                                      \
+    else if (relu)                                                           \
       bn_reduce_bwd_kernel<T, true><<<sh.rgrid, kBlock, sh.lds, s>>>(        \
           tp<T>(dy), tp<T>(x), tp<T>(y), save_mean.data_ptr<float>(),        \
           save_invstd.data_ptr<float>(), sums.data_ptr<float>(), sh.R,       \

@@ -1,0 +1,274 @@
+// LayerNorm fwd/bwd over the last dim + erf-GELU fwd/bwd (BERT path).
+//
+// Replaces the framework LayerNorm/GELU kernels the reference's stack
+// would use for a transformer (SURVEY.md §3.2 "framework elementwise",
+// BASELINE config #4).  x: [R, D] bf16/fp32, weight/bias fp32, fp32
+// statistics; one 256-thread block per row, vectorized loads.
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include "dlx_common.h"
+
+namespace dlx {
+
+namespace {
+
+struct SumR { DLX_DEV float operator()(float a, float b) const { return a + b; } };
+
+template <typename T>
+__global__ void ln_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
+                              const float* __restrict__ w,
+                              const float* __restrict__ b,
+                              float* __restrict__ save_mean,
+                              float* __restrict__ save_invstd,
+                              long R, int D, float eps) {
+  __shared__ float scratch[kBlock / kWave];
+  const long row = blockIdx.x;
+  if (row >= R) return;
+  const T* in = x + row * D;
+  T* out = y + row * D;
+  float s = 0.f, sq = 0.f;
+  for (int d = threadIdx.x; d < D; d += blockDim.x) {
+    float v = VecTraits<T>::load1(in + d);
+    s += v;
+    sq += v * v;
+  }
+  s = block_reduce(s, scratch, 0.f, SumR{});
+  __syncthreads();
+  sq = block_reduce(sq, scratch, 0.f, SumR{});
+  const float mean = s / D;
+  const float var = fmaxf(sq / D - mean * mean, 0.f);
+  const float invstd = rsqrtf(var + eps);
+  if (threadIdx.x == 0) {
+    save_mean[row] = mean;
+    save_invstd[row] = invstd;
+  }
+  for (int d = threadIdx.x; d < D; d += blockDim.x) {
+    float v = (VecTraits<T>::load1(in + d) - mean) * invstd;
+    VecTraits<T>::store1(out + d, v * w[d] + b[d]);
+  }
+}
+
+// dx = invstd * (dyw - mean(dyw) - xhat * mean(dyw * xhat)),  dyw = dy*w
+template <typename T>
+__global__ void ln_bwd_kernel(const T* __restrict__ dy,
+                              const T* __restrict__ x,
+                              const float* __restrict__ w,
+                              const float* __restrict__ save_mean,
+                              const float* __restrict__ save_invstd,
+                              T* __restrict__ dx,
+                              long R, int D) {
+  __shared__ float scratch[kBlock / kWave];
+  const long row = blockIdx.x;
+  if (row >= R) return;
+  const T* din = dy + row * D;
+  const T* xin = x + row * D;
+  const float mean = save_mean[row], invstd = save_invstd[row];
+  float s1 = 0.f, s2 = 0.f;
+  for (int d = threadIdx.x; d < D; d += blockDim.x) {
+    float g = VecTraits<T>::load1(din + d);
+    float xh = (VecTraits<T>::load1(xin + d) - mean) * invstd;
+    float gw = g * w[d];
+    s1 += gw;
+    s2 += gw * xh;
+  }
+  s1 = block_reduce(s1, scratch, 0.f, SumR{});
+  __syncthreads();
+  s2 = block_reduce(s2, scratch, 0.f, SumR{});
+  const float k1 = s1 / D, k2 = s2 / D;
+  for (int d = threadIdx.x; d < D; d += blockDim.x) {
+    float g = VecTraits<T>::load1(din + d) * w[d];
+    float xh = (VecTraits<T>::load1(xin + d) - mean) * invstd;
+    VecTraits<T>::store1(dx + row * D + d, invstd * (g - k1 - xh * k2));
+  }
+}
+
+// column-reduce for dgamma/dbeta (BN-reduce pattern: channel-tile x
+// row-stripe blocks, LDS partials, one global atomic per channel/block)
+template <typename T>
+__global__ void ln_param_kernel(const T* __restrict__ dy,
+                                const T* __restrict__ x,
+                                const float* __restrict__ save_mean,
+                                const float* __restrict__ save_invstd,
+                                float* __restrict__ dgamma,
+                                float* __restrict__ dbeta,
+                                long R, int D, int CT) {
+  extern __shared__ float lds[];
+  float* lg = lds;
+  float* lb = lds + CT;
+  for (int i = threadIdx.x; i < 2 * CT; i += blockDim.x) lds[i] = 0.f;
+  __syncthreads();
+  const int d = blockIdx.y * CT + (threadIdx.x % CT);
+  const int sub = threadIdx.x / CT;
+  const int rows_grp = blockDim.x / CT;
+  float sg = 0.f, sb = 0.f;
+  for (long r = (long)blockIdx.x * rows_grp + sub; r < R;
+       r += (long)gridDim.x * rows_grp) {
+    float g = VecTraits<T>::load1(dy + r * D + d);
+    float xh = (VecTraits<T>::load1(x + r * D + d) - save_mean[r]) *
+               save_invstd[r];
+    sg += g * xh;
+    sb += g;
+  }
+  atomicAdd(&lg[threadIdx.x % CT], sg);
+  atomicAdd(&lb[threadIdx.x % CT], sb);
+  __syncthreads();
+  if (threadIdx.x < CT) {
+    atomicAdd(&dgamma[blockIdx.y * CT + threadIdx.x], lg[threadIdx.x]);
+    atomicAdd(&dbeta[blockIdx.y * CT + threadIdx.x], lb[threadIdx.x]);
+  }
+}
+
+// ---- erf GELU ----------------------------------------------------------
+
+struct GeluOp {
+  DLX_DEV float operator()(float x) const {
+    return 0.5f * x * (1.f + erff(x * 0.70710678118654752f));
+  }
+};
+// d/dx gelu = Phi(x) + x * phi(x)
+struct GeluBwdOp {
+  DLX_DEV float operator()(float g, float x) const {
+    const float cdf = 0.5f * (1.f + erff(x * 0.70710678118654752f));
+    const float pdf = 0.3989422804014327f * __expf(-0.5f * x * x);
+    return g * (cdf + x * pdf);
+  }
+};
+
+template <typename T, typename Op>
+__global__ void gelu1_kernel(const T* a, T* o, long n, Op op) {
+  using VT = VecTraits<T>;
+  using Vec = typename VT::Vec;
+  const long nvec = n / VT::width;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < nvec;
+       i += stride) {
+    Vec x = reinterpret_cast<const Vec*>(a)[i];
+    Vec y;
+#pragma unroll
+    for (int j = 0; j < VT::width; ++j) VT::store(y, j, op(VT::load(x, j)));
+    reinterpret_cast<Vec*>(o)[i] = y;
+  }
+  const long t0 = nvec * VT::width;
+  for (long i = t0 + blockIdx.x * (long)blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    VT::store1(o + i, op(VT::load1(a + i)));
+}
+
+template <typename T>
+__global__ void gelu_bwd_kernel(const T* g, const T* x, T* o, long n) {
+  GeluBwdOp op;
+  const long stride = (long)gridDim.x * blockDim.x;
+  for (long i = blockIdx.x * (long)blockDim.x + threadIdx.x; i < n;
+       i += stride)
+    VecTraits<T>::store1(o + i, op(VecTraits<T>::load1(g + i),
+                                   VecTraits<T>::load1(x + i)));
+}
+
+}  // namespace
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> ln_fwd_impl(
+    const at::Tensor& x_in, const at::Tensor& w, const at::Tensor& b,
+    double eps) {
+  auto x = x_in.contiguous();
+  const int D = x.size(-1);
+  const long R = x.numel() / D;
+  TORCH_CHECK(w.scalar_type() == at::kFloat, "ln: params fp32");
+  auto y = at::empty_like(x);
+  auto fopts = x.options().dtype(at::kFloat);
+  auto mean = at::empty({R}, fopts);
+  auto invstd = at::empty({R}, fopts);
+  auto s = at::cuda::getCurrentCUDAStream();
+  if (x.scalar_type() == at::kBFloat16)
+    ln_fwd_kernel<uint16_t><<<R, kBlock, 0, s>>>(
+        reinterpret_cast<const uint16_t*>(x.data_ptr()),
+        reinterpret_cast<uint16_t*>(y.data_ptr()), w.data_ptr<float>(),
+        b.data_ptr<float>(), mean.data_ptr<float>(),
+        invstd.data_ptr<float>(), R, D, (float)eps);
+  else
+    ln_fwd_kernel<float><<<R, kBlock, 0, s>>>(
+        x.data_ptr<float>(), y.data_ptr<float>(), w.data_ptr<float>(),
+        b.data_ptr<float>(), mean.data_ptr<float>(),
+        invstd.data_ptr<float>(), R, D, (float)eps);
+  return {y, mean, invstd};
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> ln_bwd_impl(
+    const at::Tensor& dy_in, const at::Tensor& x_in, const at::Tensor& w,
+    const at::Tensor& mean, const at::Tensor& invstd) {
+  auto x = x_in.contiguous();
+  auto dy = dy_in.contiguous();
+  const int D = x.size(-1);
+  const long R = x.numel() / D;
+  auto dx = at::empty_like(x);
+  auto fopts = x.options().dtype(at::kFloat);
+  auto dgamma = at::zeros({D}, fopts);
+  auto dbeta = at::zeros({D}, fopts);
+  auto s = at::cuda::getCurrentCUDAStream();
+  int CT = 256;
+  while (CT > 1 && (D % CT)) CT >>= 1;
+  int rows_grp = kBlock / CT;
+  long stripes = (R + rows_grp - 1) / rows_grp;
+  int max_stripes = 2048 / std::max(1, D / CT);
+  if (stripes > max_stripes) stripes = max_stripes;
+  if (stripes < 1) stripes = 1;
+  dim3 pgrid((unsigned)stripes, (unsigned)(D / CT));
+  int lds_bytes = 2 * CT * sizeof(float);
+  if (x.scalar_type() == at::kBFloat16) {
+    ln_bwd_kernel<uint16_t><<<R, kBlock, 0, s>>>(
+        reinterpret_cast<const uint16_t*>(dy.data_ptr()),
+        reinterpret_cast<const uint16_t*>(x.data_ptr()),
+        w.data_ptr<float>(), mean.data_ptr<float>(),
+        invstd.data_ptr<float>(),
+        reinterpret_cast<uint16_t*>(dx.data_ptr()), R, D);
+    ln_param_kernel<uint16_t><<<pgrid, kBlock, lds_bytes, s>>>(
+        reinterpret_cast<const uint16_t*>(dy.data_ptr()),
+        reinterpret_cast<const uint16_t*>(x.data_ptr()),
+        mean.data_ptr<float>(), invstd.data_ptr<float>(),
+        dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), R, D, CT);
+  } else {
+    ln_bwd_kernel<float><<<R, kBlock, 0, s>>>(
+        dy.data_ptr<float>(), x.data_ptr<float>(), w.data_ptr<float>(),
+        mean.data_ptr<float>(), invstd.data_ptr<float>(),
+        dx.data_ptr<float>(), R, D);
+    ln_param_kernel<float><<<pgrid, kBlock, lds_bytes, s>>>(
+        dy.data_ptr<float>(), x.data_ptr<float>(),
+        mean.data_ptr<float>(), invstd.data_ptr<float>(),
+        dgamma.data_ptr<float>(), dbeta.data_ptr<float>(), R, D, CT);
+  }
+  return {dx, dgamma, dbeta};
+}
+
+at::Tensor gelu_fwd_impl(const at::Tensor& x_in) {
+  auto x = x_in.contiguous();
+  auto y = at::empty_like(x);
+  long n = x.numel();
+  auto s = at::cuda::getCurrentCUDAStream();
+  if (x.scalar_type() == at::kBFloat16)
+    gelu1_kernel<uint16_t, GeluOp><<<grid_for(n / 8 + 1), kBlock, 0, s>>>(
+        reinterpret_cast<const uint16_t*>(x.data_ptr()),
+        reinterpret_cast<uint16_t*>(y.data_ptr()), n, GeluOp{});
+  else
+    gelu1_kernel<float, GeluOp><<<grid_for(n / 4 + 1), kBlock, 0, s>>>(
+        x.data_ptr<float>(), y.data_ptr<float>(), n, GeluOp{});
+  return y;
+}
+
+at::Tensor gelu_bwd_impl(const at::Tensor& g_in, const at::Tensor& x_in) {
+  auto x = x_in.contiguous();
+  auto g = g_in.contiguous();
+  auto dx = at::empty_like(x);
+  long n = x.numel();
+  auto s = at::cuda::getCurrentCUDAStream();
+  if (x.scalar_type() == at::kBFloat16)
+    gelu_bwd_kernel<uint16_t><<<grid_for(n), kBlock, 0, s>>>(
+        reinterpret_cast<const uint16_t*>(g.data_ptr()),
+        reinterpret_cast<const uint16_t*>(x.data_ptr()),
+        reinterpret_cast<uint16_t*>(dx.data_ptr()), n);
+  else
+    gelu_bwd_kernel<float><<<grid_for(n), kBlock, 0, s>>>(
+        g.data_ptr<float>(), x.data_ptr<float>(), dx.data_ptr<float>(), n);
+  return dx;
+}
+
+}  // namespace dlx

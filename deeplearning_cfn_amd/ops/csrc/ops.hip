@@ -50,6 +50,27 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_bwd_impl(
     const at::Tensor& save_mean, const at::Tensor& save_invstd,
     const at::Tensor& y, bool relu);
 
+// pooling.hip
+std::tuple<at::Tensor, at::Tensor> maxpool3x3s2_fwd_impl(const at::Tensor& x);
+at::Tensor maxpool3x3s2_bwd_impl(const at::Tensor& dy, const at::Tensor& argmax,
+                                 int64_t H, int64_t W);
+at::Tensor gap_fwd_impl(const at::Tensor& x);
+at::Tensor gap_bwd_impl(const at::Tensor& dy, int64_t H, int64_t W);
+
+// pack.hip
+at::Tensor pack_nchw_to_nhwc_impl(const at::Tensor& x);
+at::Tensor pack_nhwc_to_nchw_impl(const at::Tensor& x);
+at::Tensor transpose2d_impl(const at::Tensor& x);
+
+// layernorm.hip
+std::tuple<at::Tensor, at::Tensor, at::Tensor> ln_fwd_impl(
+    const at::Tensor& x, const at::Tensor& w, const at::Tensor& b, double eps);
+std::tuple<at::Tensor, at::Tensor, at::Tensor> ln_bwd_impl(
+    const at::Tensor& dy, const at::Tensor& x, const at::Tensor& w,
+    const at::Tensor& mean, const at::Tensor& invstd);
+at::Tensor gelu_fwd_impl(const at::Tensor& x);
+at::Tensor gelu_bwd_impl(const at::Tensor& g, const at::Tensor& x);
+
 namespace {
 
 hipStream_t cur_stream() { return at::cuda::getCurrentCUDAStream(); }
@@ -195,6 +216,17 @@ TORCH_LIBRARY(dlx_hip, m) {
   m.def("gemm_nt(Tensor a, Tensor b) -> Tensor");
   m.def("gemm_nn(Tensor g, Tensor w) -> Tensor");
   m.def("gemm_tn(Tensor g, Tensor x) -> Tensor");
+  m.def("maxpool3x3s2_fwd(Tensor x) -> (Tensor, Tensor)");
+  m.def("maxpool3x3s2_bwd(Tensor dy, Tensor argmax, int H, int W) -> Tensor");
+  m.def("gap_fwd(Tensor x) -> Tensor");
+  m.def("gap_bwd(Tensor dy, int H, int W) -> Tensor");
+  m.def("pack_nchw_to_nhwc(Tensor x) -> Tensor");
+  m.def("pack_nhwc_to_nchw(Tensor x) -> Tensor");
+  m.def("transpose2d(Tensor x) -> Tensor");
+  m.def("ln_fwd(Tensor x, Tensor w, Tensor b, float eps) -> (Tensor, Tensor, Tensor)");
+  m.def("ln_bwd(Tensor dy, Tensor x, Tensor w, Tensor mean, Tensor invstd) -> (Tensor, Tensor, Tensor)");
+  m.def("gelu_fwd(Tensor x) -> Tensor");
+  m.def("gelu_bwd(Tensor g, Tensor x) -> Tensor");
 }
 
 TORCH_LIBRARY_IMPL(dlx_hip, CUDA, m) {
@@ -213,4 +245,15 @@ TORCH_LIBRARY_IMPL(dlx_hip, CUDA, m) {
   m.impl("gemm_nt", dlx::gemm_nt);
   m.impl("gemm_nn", dlx::gemm_nn);
   m.impl("gemm_tn", dlx::gemm_tn);
+  m.impl("maxpool3x3s2_fwd", dlx::maxpool3x3s2_fwd_impl);
+  m.impl("maxpool3x3s2_bwd", dlx::maxpool3x3s2_bwd_impl);
+  m.impl("gap_fwd", dlx::gap_fwd_impl);
+  m.impl("gap_bwd", dlx::gap_bwd_impl);
+  m.impl("pack_nchw_to_nhwc", dlx::pack_nchw_to_nhwc_impl);
+  m.impl("pack_nhwc_to_nchw", dlx::pack_nhwc_to_nchw_impl);
+  m.impl("transpose2d", dlx::transpose2d_impl);
+  m.impl("ln_fwd", dlx::ln_fwd_impl);
+  m.impl("ln_bwd", dlx::ln_bwd_impl);
+  m.impl("gelu_fwd", dlx::gelu_fwd_impl);
+  m.impl("gelu_bwd", dlx::gelu_bwd_impl);
 }

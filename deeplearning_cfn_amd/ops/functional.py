@@ -156,6 +156,90 @@ def nhwc_to_nchw(x: torch.Tensor) -> torch.Tensor:
 
 
 # ---------------------------------------------------------------------------
+# pooling (ResNet stem maxpool 3x3/s2/p1, global average pool)
+# ---------------------------------------------------------------------------
+
+class _MaxPool3x3s2(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        y, argmax = native().maxpool3x3s2_fwd(x)
+        ctx.save_for_backward(argmax)
+        ctx.hw = (x.shape[2], x.shape[3])
+        return y
+
+    @staticmethod
+    def backward(ctx, grad):
+        (argmax,) = ctx.saved_tensors
+        return native().maxpool3x3s2_bwd(grad, argmax, *ctx.hw)
+
+
+def maxpool3x3s2(x: torch.Tensor) -> torch.Tensor:
+    if _use_native(x):
+        return _MaxPool3x3s2.apply(x)
+    return F.max_pool2d(x, 3, stride=2, padding=1)
+
+
+class _GlobalAvgPool(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ctx.hw = (x.shape[2], x.shape[3])
+        return native().gap_fwd(x)
+
+    @staticmethod
+    def backward(ctx, grad):
+        return native().gap_bwd(grad, *ctx.hw)
+
+
+def global_avg_pool(x: torch.Tensor) -> torch.Tensor:
+    if _use_native(x):
+        return _GlobalAvgPool.apply(x)
+    return F.adaptive_avg_pool2d(x, 1)
+
+
+# ---------------------------------------------------------------------------
+# LayerNorm + GELU (BERT path)
+# ---------------------------------------------------------------------------
+
+class _LayerNorm(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias, eps):
+        y, mean, invstd = native().ln_fwd(x, weight, bias, eps)
+        ctx.save_for_backward(x, weight, mean, invstd)
+        return y
+
+    @staticmethod
+    def backward(ctx, grad):
+        x, weight, mean, invstd = ctx.saved_tensors
+        dx, dw, db = native().ln_bwd(grad, x, weight, mean, invstd)
+        return dx, dw, db, None
+
+
+def layer_norm(x, normalized_shape, weight, bias, eps=1e-5):
+    if (_use_native(x) and weight is not None
+            and weight.dtype == torch.float32 and len(normalized_shape) == 1):
+        return _LayerNorm.apply(x, weight, bias, eps)
+    return F.layer_norm(x, normalized_shape, weight, bias, eps)
+
+
+class _GELU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x):
+        ctx.save_for_backward(x)
+        return native().gelu_fwd(x)
+
+    @staticmethod
+    def backward(ctx, grad):
+        (x,) = ctx.saved_tensors
+        return native().gelu_bwd(grad, x)
+
+
+def gelu(x: torch.Tensor) -> torch.Tensor:
+    if _use_native(x):
+        return _GELU.apply(x)
+    return F.gelu(x)
+
+
+# ---------------------------------------------------------------------------
 # GEMM / Linear (MFMA kernel; used by the FC head and the BERT path)
 # ---------------------------------------------------------------------------
 
