@@ -25,12 +25,23 @@ import torch.nn as nn
 from ..ops import functional as F_amd
 
 
+class DLXConv2d(nn.Conv2d):
+    """Conv2d dispatching to the implicit-GEMM MFMA kernel (ops/conv.hip).
+
+    Falls back to the library path for shapes outside the native surface
+    (Cin%8!=0, e.g. the 3-channel stem) or when DLX_CONV=torch.
+    """
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return F_amd.conv2d(x, self.weight, self.stride[0], self.padding[0])
+
+
 def conv3x3(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
-    return nn.Conv2d(cin, cout, 3, stride=stride, padding=1, bias=False)
+    return DLXConv2d(cin, cout, 3, stride=stride, padding=1, bias=False)
 
 
 def conv1x1(cin: int, cout: int, stride: int = 1) -> nn.Conv2d:
-    return nn.Conv2d(cin, cout, 1, stride=stride, bias=False)
+    return DLXConv2d(cin, cout, 1, stride=stride, bias=False)
 
 
 class StemMaxPool(nn.Module):
@@ -107,7 +118,7 @@ class ResNet(nn.Module):
         self.stem_kind = stem
         self.inplanes = width if stem == "imagenet" else 16
         if stem == "imagenet":
-            self.conv1 = nn.Conv2d(3, self.inplanes, 7, stride=2, padding=3,
+            self.conv1 = DLXConv2d(3, self.inplanes, 7, stride=2, padding=3,
                                    bias=False)
             self.maxpool = StemMaxPool()
         else:  # cifar: single 3x3 stem, no maxpool

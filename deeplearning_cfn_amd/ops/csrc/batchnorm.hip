@@ -63,9 +63,10 @@ __global__ void bn_reduce_fwd_vec_kernel(const T* __restrict__ x,
     atomicAdd(&lsq[j], sq[j]);
   }
   __syncthreads();
+  float* part = sums + (long)blockIdx.x * 2 * C;
   for (int i = threadIdx.x; i < CT * VW; i += blockDim.x) {
-    atomicAdd(&sums[blockIdx.y * CT * VW + i], lds[i]);
-    atomicAdd(&sums[C + blockIdx.y * CT * VW + i], lds[CT * VW + i]);
+    part[blockIdx.y * CT * VW + i] = lds[i];
+    part[C + blockIdx.y * CT * VW + i] = lds[CT * VW + i];
   }
 }
 
@@ -92,9 +93,10 @@ __global__ void bn_reduce_fwd_kernel(const T* __restrict__ x,
   atomicAdd(&lsum[threadIdx.x % CT], s);
   atomicAdd(&lsq[threadIdx.x % CT], sq);
   __syncthreads();
+  float* part = sums + (long)blockIdx.x * 2 * C;
   if (threadIdx.x < CT) {
-    atomicAdd(&sums[blockIdx.y * CT + threadIdx.x], lsum[threadIdx.x]);
-    atomicAdd(&sums[C + blockIdx.y * CT + threadIdx.x], lsq[threadIdx.x]);
+    part[blockIdx.y * CT + threadIdx.x] = lsum[threadIdx.x];
+    part[C + blockIdx.y * CT + threadIdx.x] = lsq[threadIdx.x];
   }
 }
 
@@ -109,12 +111,17 @@ __global__ void bn_finalize_fwd_kernel(const float* __restrict__ sums,
                                        float* __restrict__ save_invstd,
                                        float* __restrict__ scale,
                                        float* __restrict__ shift,
-                                       long R, int C, float momentum,
-                                       float eps) {
+                                       long R, int C, int nstripes,
+                                       float momentum, float eps) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  float mean = sums[c] / R;
-  float var = fmaxf(sums[C + c] / R - mean * mean, 0.f);  // biased
+  float s1 = 0.f, s2 = 0.f;
+  for (int st = 0; st < nstripes; ++st) {
+    s1 += sums[(long)st * 2 * C + c];
+    s2 += sums[(long)st * 2 * C + C + c];
+  }
+  float mean = s1 / R;
+  float var = fmaxf(s2 / R - mean * mean, 0.f);  // biased
   float invstd = rsqrtf(var + eps);
   save_mean[c] = mean;
   save_invstd[c] = invstd;
@@ -229,9 +236,10 @@ __global__ void bn_reduce_bwd_vec_kernel(const T* __restrict__ dy,
     atomicAdd(&s_dyx[j], bx[j]);
   }
   __syncthreads();
+  float* part = sums + (long)blockIdx.x * 2 * C;
   for (int i = threadIdx.x; i < CT * VW; i += blockDim.x) {
-    atomicAdd(&sums[blockIdx.y * CT * VW + i], lds[i]);
-    atomicAdd(&sums[C + blockIdx.y * CT * VW + i], lds[CT * VW + i]);
+    part[blockIdx.y * CT * VW + i] = lds[i];
+    part[C + blockIdx.y * CT * VW + i] = lds[CT * VW + i];
   }
 }
 
@@ -266,9 +274,10 @@ __global__ void bn_reduce_bwd_kernel(const T* __restrict__ dy,
   atomicAdd(&s_dy[threadIdx.x % CT], a);
   atomicAdd(&s_dyx[threadIdx.x % CT], bx);
   __syncthreads();
+  float* part = sums + (long)blockIdx.x * 2 * C;
   if (threadIdx.x < CT) {
-    atomicAdd(&sums[blockIdx.y * CT + threadIdx.x], s_dy[threadIdx.x]);
-    atomicAdd(&sums[C + blockIdx.y * CT + threadIdx.x], s_dyx[threadIdx.x]);
+    part[blockIdx.y * CT + threadIdx.x] = s_dy[threadIdx.x];
+    part[C + blockIdx.y * CT + threadIdx.x] = s_dyx[threadIdx.x];
   }
 }
 
@@ -280,13 +289,18 @@ __global__ void bn_finalize_bwd_kernel(const float* __restrict__ sums,
                                        float* __restrict__ k1,  // sum_dy/R
                                        float* __restrict__ k2,  // sum_dyxhat/R
                                        float* __restrict__ gi,  // gamma*invstd
-                                       long R, int C) {
+                                       long R, int C, int nstripes) {
   int c = blockIdx.x * blockDim.x + threadIdx.x;
   if (c >= C) return;
-  dbeta[c] = sums[c];
-  dgamma[c] = sums[C + c];
-  k1[c] = sums[c] / R;
-  k2[c] = sums[C + c] / R;
+  float s1 = 0.f, s2 = 0.f;
+  for (int st = 0; st < nstripes; ++st) {
+    s1 += sums[(long)st * 2 * C + c];
+    s2 += sums[(long)st * 2 * C + C + c];
+  }
+  dbeta[c] = s1;
+  dgamma[c] = s2;
+  k1[c] = s1 / R;
+  k2[c] = s2 / R;
   gi[c] = w[c] * save_invstd[c];
 }
 
@@ -389,7 +403,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd_train_impl(
   TORCH_CHECK(bf || x.scalar_type() == at::kFloat, "bn: bf16/fp32 only");
   auto sh = bn_shape(x, bf ? 8 : 4);
   auto fopts = x.options().dtype(at::kFloat);
-  auto sums = at::zeros({2 * sh.C}, fopts);
+  auto sums = at::empty({(long)sh.rgrid.x * 2 * sh.C}, fopts);
   auto save_mean = at::empty({sh.C}, fopts);
   auto save_invstd = at::empty({sh.C}, fopts);
   auto scale = at::empty({sh.C}, fopts);
@@ -410,7 +424,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd_train_impl(
       rmean.data_ptr<float>(), rvar.data_ptr<float>(),                       \
       save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),            \
       scale.data_ptr<float>(), shift.data_ptr<float>(), sh.R, sh.C,          \
-      (float)momentum, (float)eps);                                          \
+      (int)sh.rgrid.x, (float)momentum, (float)eps);                         \
   {                                                                          \
     const bool vec = (sh.C % VecTraits<T>::width) == 0;                      \
     int grid = grid_for(total / (vec ? VecTraits<T>::width : 1) + 1);        \
@@ -491,7 +505,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_bwd_impl(
   const bool bf = x.scalar_type() == at::kBFloat16;
   auto sh = bn_shape(x, bf ? 8 : 4);
   auto fopts = x.options().dtype(at::kFloat);
-  auto sums = at::zeros({2 * sh.C}, fopts);
+  auto sums = at::empty({(long)sh.rgrid.x * 2 * sh.C}, fopts);
   auto dgamma = at::empty({sh.C}, fopts);
   auto dbeta = at::empty({sh.C}, fopts);
   auto k1 = at::empty({sh.C}, fopts);
@@ -526,7 +540,8 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_bwd_impl(
         sums.data_ptr<float>(), w.data_ptr<float>(),                         \
         save_invstd.data_ptr<float>(), dgamma.data_ptr<float>(),             \
         dbeta.data_ptr<float>(), k1.data_ptr<float>(),                       \
-        k2.data_ptr<float>(), gi.data_ptr<float>(), sh.R, sh.C);             \
+        k2.data_ptr<float>(), gi.data_ptr<float>(), sh.R, sh.C,              \
+        (int)sh.rgrid.x);                                                    \
     const bool vec = (sh.C % VecTraits<T>::width) == 0;                      \
     int grid = grid_for(total / (vec ? VecTraits<T>::width : 1) + 1);        \
     if (relu && vec)                                                         \

@@ -62,6 +62,13 @@ at::Tensor pack_nchw_to_nhwc_impl(const at::Tensor& x);
 at::Tensor pack_nhwc_to_nchw_impl(const at::Tensor& x);
 at::Tensor transpose2d_impl(const at::Tensor& x);
 
+// conv.hip
+at::Tensor conv2d_fwd_impl(const at::Tensor& x, const at::Tensor& w,
+                           int64_t stride, int64_t pad);
+at::Tensor conv2d_dgrad_impl(const at::Tensor& g, const at::Tensor& w,
+                             int64_t H, int64_t W, int64_t stride,
+                             int64_t pad);
+
 // layernorm.hip
 std::tuple<at::Tensor, at::Tensor, at::Tensor> ln_fwd_impl(
     const at::Tensor& x, const at::Tensor& w, const at::Tensor& b, double eps);
@@ -223,6 +230,8 @@ TORCH_LIBRARY(dlx_hip, m) {
   m.def("pack_nchw_to_nhwc(Tensor x) -> Tensor");
   m.def("pack_nhwc_to_nchw(Tensor x) -> Tensor");
   m.def("transpose2d(Tensor x) -> Tensor");
+  m.def("conv2d_fwd(Tensor x, Tensor w, int stride, int pad) -> Tensor");
+  m.def("conv2d_dgrad(Tensor g, Tensor w, int H, int W, int stride, int pad) -> Tensor");
   m.def("ln_fwd(Tensor x, Tensor w, Tensor b, float eps) -> (Tensor, Tensor, Tensor)");
   m.def("ln_bwd(Tensor dy, Tensor x, Tensor w, Tensor mean, Tensor invstd) -> (Tensor, Tensor, Tensor)");
   m.def("gelu_fwd(Tensor x) -> Tensor");
@@ -252,6 +261,8 @@ TORCH_LIBRARY_IMPL(dlx_hip, CUDA, m) {
   m.impl("pack_nchw_to_nhwc", dlx::pack_nchw_to_nhwc_impl);
   m.impl("pack_nhwc_to_nchw", dlx::pack_nhwc_to_nchw_impl);
   m.impl("transpose2d", dlx::transpose2d_impl);
+  m.impl("conv2d_fwd", dlx::conv2d_fwd_impl);
+  m.impl("conv2d_dgrad", dlx::conv2d_dgrad_impl);
   m.impl("ln_fwd", dlx::ln_fwd_impl);
   m.impl("ln_bwd", dlx::ln_bwd_impl);
   m.impl("gelu_fwd", dlx::gelu_fwd_impl);

@@ -156,6 +156,55 @@ def nhwc_to_nchw(x: torch.Tensor) -> torch.Tensor:
 
 
 # ---------------------------------------------------------------------------
+# convolution — implicit-GEMM MFMA (fwd + dgrad native; wgrad library)
+# ---------------------------------------------------------------------------
+
+import os as _os
+
+
+def _conv_policy() -> str:
+    return _os.environ.get("DLX_CONV", "native")
+
+
+class _Conv2d(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, stride: int, pad: int):
+        y = native().conv2d_fwd(x, weight, stride, pad)
+        ctx.save_for_backward(x, weight)
+        ctx.sp = (stride, pad)
+        return y
+
+    @staticmethod
+    def backward(ctx, grad):
+        x, w = ctx.saved_tensors
+        s, p = ctx.sp
+        grad = grad.contiguous(memory_format=torch.channels_last)
+        dx = dw = None
+        if ctx.needs_input_grad[0]:
+            dx = native().conv2d_dgrad(grad, w, x.shape[2], x.shape[3], s, p)
+        if ctx.needs_input_grad[1]:
+            # wgrad stays on the library path until the rr-GEMM kernel lands
+            dw = torch.ops.aten.convolution_backward(
+                grad, x, w, None, [s, s], [p, p], [1, 1], False, [0, 0], 1,
+                [False, True, False])[1]
+        return dx, dw, None, None
+
+
+def conv2d(x: torch.Tensor, weight: torch.Tensor, stride: int = 1,
+           padding: int = 0) -> torch.Tensor:
+    """2D conv, no bias/dilation/groups (the ResNet surface).
+
+    Native implicit-GEMM when: GPU + bf16 + Cin%8==0 + Cout%8==0 and the
+    policy allows; otherwise the library path (stem conv Cin=3 included).
+    """
+    if (_use_native(x) and _conv_policy() == "native"
+            and x.dtype == torch.bfloat16
+            and weight.shape[1] % 8 == 0 and weight.shape[0] % 8 == 0):
+        return _Conv2d.apply(x, weight, stride, padding)
+    return F.conv2d(x, weight, None, stride, padding)
+
+
+# ---------------------------------------------------------------------------
 # pooling (ResNet stem maxpool 3x3/s2/p1, global average pool)
 # ---------------------------------------------------------------------------
 
