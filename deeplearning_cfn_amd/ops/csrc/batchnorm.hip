@@ -371,8 +371,9 @@ static BNShape bn_shape(const at::Tensor& x, int vw) {
   s.CT = channel_tile(units);
   int rows_grp = kBlock / s.CT;
   long stripes = (s.R + rows_grp - 1) / rows_grp;
-  // enough blocks to fill 256 CUs x 8 XCDs without atomics blowup
-  int max_stripes = 2048 / std::max(1, units / s.CT);
+  // ~512 blocks total fills the 256 CUs ~2x; more stripes only grows the
+  // partial buffer and the finalize loop
+  int max_stripes = std::max(1, 512 / std::max(1, units / s.CT));
   if (stripes > max_stripes) stripes = max_stripes;
   if (stripes < 1) stripes = 1;
   s.rgrid = dim3((unsigned)stripes, (unsigned)(units / s.CT));

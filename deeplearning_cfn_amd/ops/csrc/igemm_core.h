@@ -1,0 +1,226 @@
+// Unified MFMA implicit-GEMM core (gfx950).
+//
+// One templated kernel serves plain GEMM (C = A·B^T, both K-contiguous)
+// and NHWC convolution fwd/dgrad (the im2col gather fused into staging):
+// every operand element is fetched by global_load_lds with a PER-LANE
+// source address (guide §5: the glds source is per-lane; only the LDS
+// destination must be lane-linear), and any out-of-range tap (padding,
+// M/N/K tails) redirects to a 16-B zero page — no separate edge kernels.
+//
+// LDS layout per operand tile: [BK/8][ROWS][bf16x8] — k-subtile-major, so
+//  * glds destination is lane-linear (slot f = kc*ROWS + row), and
+//  * fragment ds_read_b128 is conflict-free: each 16-lane group reads 16
+//    CONSECUTIVE 16-B slots (one full 256-B bank row).
+//
+// Geometries: 128x128 (4 waves, 2x2) for narrow problems; 256x256
+// (8 waves, 2x4, 128 KiB LDS, 1 block/CU) for the big ones.
+// Double-buffered, stage(t+1) issued before compute(t), one
+// vmcnt(0)+barrier per K-tile (guide §5.5 T3 minimum 2-phase).
+#pragma once
+
+#include "dlx_common.h"
+
+namespace dlx {
+
+using bf16x8 = __attribute__((ext_vector_type(8))) short;
+using f32x4 = __attribute__((ext_vector_type(4))) float;
+
+typedef __attribute__((address_space(3))) uint32_t lds_u32_t;
+typedef const __attribute__((address_space(1))) uint32_t glb_u32_t;
+
+// ---- gather policies ---------------------------------------------------
+
+// plain row-major [rows, K] operand (GEMM A/B, conv weights)
+struct RowGather {
+  const uint16_t* base;
+  long rows, K;
+  DLX_DEV const uint16_t* src(long row, long k,
+                              const uint16_t* zp) const {
+    return (row < rows && k < K) ? base + row * K + k : zp;
+  }
+};
+
+struct ConvShape {
+  int N, H, W, Ci, Ho, Wo, Co, Kh, Kw, stride, pad;
+  long K, M;
+};
+
+// conv fwd activations: row m=(n,ho,wo), k=(dh,dw,ci)
+struct ConvFwdGather {
+  const uint16_t* base;
+  ConvShape p;
+  DLX_DEV const uint16_t* src(long m, long k, const uint16_t* zp) const {
+    if (m >= p.M || k >= p.K) return zp;
+    const int hw = p.Ho * p.Wo;
+    const int n = (int)(m / hw);
+    const int rem = (int)(m - (long)n * hw);
+    const int ho = rem / p.Wo, wo = rem - (rem / p.Wo) * p.Wo;
+    const int dhw = (int)(k / p.Ci);
+    const int ci = (int)(k - (long)dhw * p.Ci);
+    const int dh = dhw / p.Kw, dw = dhw - (dhw / p.Kw) * p.Kw;
+    const int hi = ho * p.stride - p.pad + dh;
+    const int wi = wo * p.stride - p.pad + dw;
+    if (hi < 0 || hi >= p.H || wi < 0 || wi >= p.W) return zp;
+    return base + (((long)n * p.H + hi) * p.W + wi) * p.Ci + ci;
+  }
+};
+
+// conv dgrad: row m=(n,hi,wi) over INPUT pixels, k=(dh,dw,co) from grad
+struct ConvDgradGather {
+  const uint16_t* base;
+  ConvShape p;
+  DLX_DEV const uint16_t* src(long m, long k, const uint16_t* zp) const {
+    if (m >= p.M || k >= p.K) return zp;
+    const int hw = p.H * p.W;
+    const int n = (int)(m / hw);
+    const int rem = (int)(m - (long)n * hw);
+    const int hi = rem / p.W, wi = rem - (rem / p.W) * p.W;
+    const int dhw = (int)(k / p.Co);
+    const int co = (int)(k - (long)dhw * p.Co);
+    const int dh = dhw / p.Kw, dw = dhw - (dhw / p.Kw) * p.Kw;
+    const int ht = hi + p.pad - dh;
+    const int wt = wi + p.pad - dw;
+    if (ht < 0 || wt < 0 || (ht % p.stride) || (wt % p.stride)) return zp;
+    const int ho = ht / p.stride, wo = wt / p.stride;
+    if (ho >= p.Ho || wo >= p.Wo) return zp;
+    return base + (((long)n * p.Ho + ho) * p.Wo + wo) * p.Co + co;
+  }
+};
+
+// ---- the kernel --------------------------------------------------------
+
+template <int BM, int BN, int BK, int WM, int WN,
+          class AG, class BG, bool OUT_BF16>
+__global__ __launch_bounds__(WM* WN * 64) void igemm_kernel(
+    AG ag, BG bg, void* __restrict__ Cout_,
+    const uint16_t* __restrict__ zero_page,
+    long M, long Nc, long K, int grid_n) {
+  constexpr int THREADS = WM * WN * 64;
+  constexpr int SLOTS_A = (BK / 8) * BM;
+  constexpr int SLOTS_B = (BK / 8) * BN;
+  constexpr int GA = SLOTS_A / THREADS;
+  constexpr int GB = SLOTS_B / THREADS;
+  constexpr int MR = (BM / WM) / 16;
+  constexpr int NR = (BN / WN) / 16;
+
+  __shared__ uint16_t lsA[2 * SLOTS_A * 8];
+  __shared__ uint16_t lsB[2 * SLOTS_B * 8];
+
+  int nwg = gridDim.x;
+  int bid = blockIdx.x;
+  {  // XCD-aware bijective remap (guide T1)
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = bid % 8, idx = bid / 8;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int bm = bid / grid_n, bn = bid % grid_n;
+  const long m0 = (long)bm * BM, n0 = (long)bn * BN;
+
+  const int lane = threadIdx.x % kWave;
+  const int wid = threadIdx.x / kWave;
+  const int wr = wid / WN, wc = wid % WN;
+  const int l15 = lane & 15, l4 = lane >> 4;
+
+  f32x4 acc[MR][NR] = {};
+  const long ntile = (K + BK - 1) / BK;
+
+  auto stage = [&](int buf, long t) {
+    const long k0 = t * BK;
+    uint16_t* la = lsA + buf * SLOTS_A * 8;
+    uint16_t* lb = lsB + buf * SLOTS_B * 8;
+#pragma unroll
+    for (int i = 0; i < GA; ++i) {
+      int f = i * THREADS + threadIdx.x;
+      int kc = f / BM;
+      int m = f - kc * BM;
+      const uint16_t* sa = ag.src(m0 + m, k0 + kc * 8, zero_page);
+      __builtin_amdgcn_global_load_lds(
+          (glb_u32_t*)sa, (lds_u32_t*)(la + (long)f * 8), 16, 0, 0);
+    }
+#pragma unroll
+    for (int i = 0; i < GB; ++i) {
+      int f = i * THREADS + threadIdx.x;
+      int kc = f / BN;
+      int n = f - kc * BN;
+      const uint16_t* sb = bg.src(n0 + n, k0 + kc * 8, zero_page);
+      __builtin_amdgcn_global_load_lds(
+          (glb_u32_t*)sb, (lds_u32_t*)(lb + (long)f * 8), 16, 0, 0);
+    }
+  };
+
+  stage(0, 0);
+  __builtin_amdgcn_s_waitcnt(0);
+  __syncthreads();
+
+  int buf = 0;
+  for (long t = 0; t < ntile; ++t) {
+    if (t + 1 < ntile) stage(buf ^ 1, t + 1);
+    const uint16_t* la = lsA + buf * SLOTS_A * 8;
+    const uint16_t* lb = lsB + buf * SLOTS_B * 8;
+#pragma unroll
+    for (int ks = 0; ks < BK / 32; ++ks) {
+      bf16x8 af[MR], bfr[NR];
+#pragma unroll
+      for (int mi = 0; mi < MR; ++mi) {
+        int m = wr * (BM / WM) + mi * 16 + l15;
+        int kc = ks * 4 + l4;
+        af[mi] = *reinterpret_cast<const bf16x8*>(la + ((long)kc * BM + m) * 8);
+      }
+#pragma unroll
+      for (int ni = 0; ni < NR; ++ni) {
+        int n = wc * (BN / WN) + ni * 16 + l15;
+        int kc = ks * 4 + l4;
+        bfr[ni] = *reinterpret_cast<const bf16x8*>(lb + ((long)kc * BN + n) * 8);
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mi = 0; mi < MR; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < NR; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    __syncthreads();
+    buf ^= 1;
+  }
+
+  // epilogue: C/D map for 16x16x32: col = lane&15, row = (lane>>4)*4 + r
+#pragma unroll
+  for (int mi = 0; mi < MR; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < NR; ++ni) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long row = m0 + wr * (BM / WM) + mi * 16 + l4 * 4 + r;
+        long col = n0 + wc * (BN / WN) + ni * 16 + l15;
+        if (row < M && col < Nc) {
+          if (OUT_BF16)
+            reinterpret_cast<uint16_t*>(Cout_)[row * Nc + col] =
+                f2bf(acc[mi][ni][r]);
+          else
+            reinterpret_cast<float*>(Cout_)[row * Nc + col] = acc[mi][ni][r];
+        }
+      }
+    }
+  }
+}
+
+// launch helper: picks 256^2 geometry for big problems, 128^2 otherwise
+template <class AG, class BG, bool OUT_BF16>
+inline void igemm_launch(const AG& ag, const BG& bg, void* cptr,
+                         const uint16_t* zp, long M, long Nc, long K,
+                         hipStream_t s) {
+  if (M >= 256 && Nc >= 192) {
+    const int gm = (int)((M + 255) / 256), gn = (int)((Nc + 255) / 256);
+    igemm_kernel<256, 256, 64, 2, 4, AG, BG, OUT_BF16>
+        <<<gm * gn, 512, 0, s>>>(ag, bg, cptr, zp, M, Nc, K, gn);
+  } else {
+    const int gm = (int)((M + 127) / 128), gn = (int)((Nc + 127) / 128);
+    igemm_kernel<128, 128, 64, 2, 2, AG, BG, OUT_BF16>
+        <<<gm * gn, 256, 0, s>>>(ag, bg, cptr, zp, M, Nc, K, gn);
+  }
+}
+
+}  // namespace dlx
