@@ -45,6 +45,7 @@ __global__ void bn_reduce_fwd_vec_kernel(const T* __restrict__ x,
   const int sub = threadIdx.x / CT;
   const int rows_grp = blockDim.x / CT;
   float s[VW] = {}, sq[VW] = {};
+#pragma unroll 4
   for (long r = (long)blockIdx.x * rows_grp + sub; r < R;
        r += (long)gridDim.x * rows_grp) {
     Vec v = *reinterpret_cast<const Vec*>(x + r * C + c0);
@@ -113,13 +114,23 @@ __global__ void bn_finalize_fwd_kernel(const float* __restrict__ sums,
                                        float* __restrict__ shift,
                                        long R, int C, int nstripes,
                                        float momentum, float eps) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  __shared__ float l1[256], l2[256];
+  const int cl = threadIdx.x & 63;          // channel lane
+  const int sl = threadIdx.x >> 6;          // stripe slice (0..3)
+  int c = blockIdx.x * 64 + cl;
   float s1 = 0.f, s2 = 0.f;
-  for (int st = 0; st < nstripes; ++st) {
-    s1 += sums[(long)st * 2 * C + c];
-    s2 += sums[(long)st * 2 * C + C + c];
+  if (c < C) {
+    for (int st = sl; st < nstripes; st += 4) {
+      s1 += sums[(long)st * 2 * C + c];
+      s2 += sums[(long)st * 2 * C + C + c];
+    }
   }
+  l1[threadIdx.x] = s1;
+  l2[threadIdx.x] = s2;
+  __syncthreads();
+  if (sl != 0 || c >= C) return;
+  s1 = l1[cl] + l1[cl + 64] + l1[cl + 128] + l1[cl + 192];
+  s2 = l2[cl] + l2[cl + 64] + l2[cl + 128] + l2[cl + 192];
   float mean = s1 / R;
   float var = fmaxf(s2 / R - mean * mean, 0.f);  // biased
   float invstd = rsqrtf(var + eps);
@@ -212,6 +223,7 @@ __global__ void bn_reduce_bwd_vec_kernel(const T* __restrict__ dy,
     invstd[j] = save_invstd[c0 + j];
   }
   float a[VW] = {}, bx[VW] = {};
+#pragma unroll 4
   for (long r = (long)blockIdx.x * rows_grp + sub; r < R;
        r += (long)gridDim.x * rows_grp) {
     const long i = r * C + c0;
@@ -290,13 +302,23 @@ __global__ void bn_finalize_bwd_kernel(const float* __restrict__ sums,
                                        float* __restrict__ k2,  // sum_dyxhat/R
                                        float* __restrict__ gi,  // gamma*invstd
                                        long R, int C, int nstripes) {
-  int c = blockIdx.x * blockDim.x + threadIdx.x;
-  if (c >= C) return;
+  __shared__ float l1[256], l2[256];
+  const int cl = threadIdx.x & 63;
+  const int sl = threadIdx.x >> 6;
+  int c = blockIdx.x * 64 + cl;
   float s1 = 0.f, s2 = 0.f;
-  for (int st = 0; st < nstripes; ++st) {
-    s1 += sums[(long)st * 2 * C + c];
-    s2 += sums[(long)st * 2 * C + C + c];
+  if (c < C) {
+    for (int st = sl; st < nstripes; st += 4) {
+      s1 += sums[(long)st * 2 * C + c];
+      s2 += sums[(long)st * 2 * C + C + c];
+    }
   }
+  l1[threadIdx.x] = s1;
+  l2[threadIdx.x] = s2;
+  __syncthreads();
+  if (sl != 0 || c >= C) return;
+  s1 = l1[cl] + l1[cl + 64] + l1[cl + 128] + l1[cl + 192];
+  s2 = l2[cl] + l2[cl + 64] + l2[cl + 128] + l2[cl + 192];
   dbeta[c] = s1;
   dgamma[c] = s2;
   k1[c] = s1 / R;
@@ -373,7 +395,7 @@ static BNShape bn_shape(const at::Tensor& x, int vw) {
   long stripes = (s.R + rows_grp - 1) / rows_grp;
   // ~512 blocks total fills the 256 CUs ~2x; more stripes only grows the
   // partial buffer and the finalize loop
-  int max_stripes = std::max(1, 512 / std::max(1, units / s.CT));
+  int max_stripes = std::max(1, 1024 / std::max(1, units / s.CT));
   if (stripes > max_stripes) stripes = max_stripes;
   if (stripes < 1) stripes = 1;
   s.rgrid = dim3((unsigned)stripes, (unsigned)(units / s.CT));
@@ -420,7 +442,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd_train_impl(
   else                                                                       \
     bn_reduce_fwd_kernel<T><<<sh.rgrid, kBlock, sh.lds, s>>>(                \
         tp<T>(x), sums.data_ptr<float>(), sh.R, sh.C, sh.CT);                \
-  bn_finalize_fwd_kernel<<<(sh.C + 255) / 256, 256, 0, s>>>(                 \
+  bn_finalize_fwd_kernel<<<(sh.C + 63) / 64, 256, 0, s>>>(                 \
       sums.data_ptr<float>(), w.data_ptr<float>(), b.data_ptr<float>(),      \
       rmean.data_ptr<float>(), rvar.data_ptr<float>(),                       \
       save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),            \
@@ -537,7 +559,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_bwd_impl(
           tp<T>(dy), tp<T>(x), tp<T>(y), save_mean.data_ptr<float>(),        \
           save_invstd.data_ptr<float>(), sums.data_ptr<float>(), sh.R,       \
           sh.C, sh.CT);                                                      \
-    bn_finalize_bwd_kernel<<<(sh.C + 255) / 256, 256, 0, s>>>(               \
+    bn_finalize_bwd_kernel<<<(sh.C + 63) / 64, 256, 0, s>>>(               \
         sums.data_ptr<float>(), w.data_ptr<float>(),                         \
         save_invstd.data_ptr<float>(), dgamma.data_ptr<float>(),             \
         dbeta.data_ptr<float>(), k1.data_ptr<float>(),                       \

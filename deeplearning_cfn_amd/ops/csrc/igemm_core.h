@@ -207,14 +207,166 @@ __global__ __launch_bounds__(WM* WN * 64) void igemm_kernel(
   }
 }
 
-// launch helper: picks 256^2 geometry for big problems, 128^2 otherwise
+// ---------------------------------------------------------------------
+// 8-phase counted-vmcnt pipeline, 256x256xBK64, 8 waves (guide §5 "256²
+// 8-phase template" adapted to the k-subtile LDS layout):
+//  * ONE __shared__ array (guide §5 trap 4a: a second LDS object makes
+//    hipcc drain vmcnt(0) before every ds_read of a glds pipeline);
+//  * stage units = (operand, k-half): 2 glds each; one unit issued per
+//    phase, so 3 units stay in flight across barriers -> vmcnt(6);
+//  * raw s_barrier (never __syncthreads: its fence drains pending
+//    LDS-DMA), each wave waits its OWN glds before the barrier;
+//  * waits only at phases 0/2 of each K-tile (k-half boundaries);
+//  * s_setprio(1) around each 16-MFMA cluster (T5 — the phase split
+//    creates the wave role diversity it needs).
+// ---------------------------------------------------------------------
+
+template <class AG, class BG, bool OUT_BF16>
+__global__ __launch_bounds__(512) void igemm_kernel_8ph(
+    AG ag, BG bg, void* __restrict__ Cout_,
+    const uint16_t* __restrict__ zero_page,
+    long M, long Nc, long K, int grid_n) {
+  constexpr int BM = 256, BN = 256, BK = 64;
+  constexpr int THREADS = 512;
+  constexpr int SLOTS = (BK / 8) * BM;          // 2048 16-B slots/operand
+  constexpr int MR = 8, NR = 4;                 // per-wave 128x64
+
+  __shared__ uint16_t lds[2 * 2 * SLOTS * 8];   // [buf][A|B][SLOTS][8]
+
+  int nwg = gridDim.x;
+  int bid = blockIdx.x;
+  {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = bid % 8, idx = bid / 8;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int bm = bid / grid_n, bn = bid % grid_n;
+  const long m0 = (long)bm * BM, n0 = (long)bn * BN;
+
+  const int lane = threadIdx.x % kWave;
+  const int wid = threadIdx.x / kWave;
+  const int wr = wid / 4, wc = wid % 4;         // 2 x 4 waves
+  const int l15 = lane & 15, l4 = lane >> 4;
+
+  f32x4 acc[MR][NR] = {};
+  const long ntile = (K + BK - 1) / BK;
+
+  // stage unit u of tile t into buffer `buf`:
+  //   u=0: A k-half 0 (glds instrs 0,1) ; u=1: B k-half 0
+  //   u=2: A k-half 1 (instrs 2,3)      ; u=3: B k-half 1
+  auto stage_unit = [&](int buf, long t, int u) {
+    const long k0 = t * BK;
+    const bool isB = u & 1;
+    const int half = u >> 1;
+    uint16_t* l0 = lds + (buf * 2 + (isB ? 1 : 0)) * SLOTS * 8;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      int f = (half * 2 + i) * THREADS + threadIdx.x;
+      int kc = f / BM;
+      int r = f - kc * BM;
+      const uint16_t* src =
+          isB ? bg.src(n0 + r, k0 + kc * 8, zero_page)
+              : ag.src(m0 + r, k0 + kc * 8, zero_page);
+      __builtin_amdgcn_global_load_lds(
+          (glb_u32_t*)src, (lds_u32_t*)(l0 + (long)f * 8), 16, 0, 0);
+    }
+  };
+
+  // compute phase q (q>>1 = ks half, q&1 = mi half) of the tile in `buf`
+  auto phase_mfma = [&](int buf, int q) {
+    const uint16_t* la = lds + (buf * 2 + 0) * SLOTS * 8;
+    const uint16_t* lb = lds + (buf * 2 + 1) * SLOTS * 8;
+    const int ks = q >> 1;
+    const int mih = (q & 1) * 4;
+    bf16x8 af[4], bfr[NR];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      int m = wr * 128 + (mih + i) * 16 + l15;
+      int kc = ks * 4 + l4;
+      af[i] = *reinterpret_cast<const bf16x8*>(la + ((long)kc * BM + m) * 8);
+    }
+#pragma unroll
+    for (int ni = 0; ni < NR; ++ni) {
+      int n = wc * 64 + ni * 16 + l15;
+      int kc = ks * 4 + l4;
+      bfr[ni] = *reinterpret_cast<const bf16x8*>(lb + ((long)kc * BN + n) * 8);
+    }
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int ni = 0; ni < NR; ++ni)
+        acc[mih + i][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[i], bfr[ni], acc[mih + i][ni], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+  };
+
+  // prologue: all 4 units of tile 0
+#pragma unroll
+  for (int u = 0; u < 4; ++u) stage_unit(0, 0, u);
+
+  int buf = 0;
+  long t = 0;
+  for (; t + 1 < ntile; ++t) {
+    // phase 0: needs A/B k-half0 of t; 3 units may remain in flight
+    stage_unit(buf ^ 1, t + 1, 0);
+    asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    phase_mfma(buf, 0);
+    // phase 1: same k-half, no new data needed
+    stage_unit(buf ^ 1, t + 1, 1);
+    __builtin_amdgcn_s_barrier();
+    phase_mfma(buf, 1);
+    // phase 2: needs k-half1 of t
+    stage_unit(buf ^ 1, t + 1, 2);
+    asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    phase_mfma(buf, 2);
+    // phase 3
+    stage_unit(buf ^ 1, t + 1, 3);
+    __builtin_amdgcn_s_barrier();
+    phase_mfma(buf, 3);
+    buf ^= 1;
+  }
+  // epilogue tile: nothing left to stage
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  phase_mfma(buf, 0);
+  phase_mfma(buf, 1);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  phase_mfma(buf, 2);
+  phase_mfma(buf, 3);
+
+#pragma unroll
+  for (int mi = 0; mi < MR; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < NR; ++ni) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        long row = m0 + wr * 128 + mi * 16 + l4 * 4 + r;
+        long col = n0 + wc * 64 + ni * 16 + l15;
+        if (row < M && col < Nc) {
+          if (OUT_BF16)
+            reinterpret_cast<uint16_t*>(Cout_)[row * Nc + col] =
+                f2bf(acc[mi][ni][r]);
+          else
+            reinterpret_cast<float*>(Cout_)[row * Nc + col] = acc[mi][ni][r];
+        }
+      }
+    }
+  }
+}
+
+// launch helper: picks the 8-phase 256^2 pipeline for big problems,
+// the simple 128^2 kernel otherwise
 template <class AG, class BG, bool OUT_BF16>
 inline void igemm_launch(const AG& ag, const BG& bg, void* cptr,
                          const uint16_t* zp, long M, long Nc, long K,
                          hipStream_t s) {
-  if (M >= 256 && Nc >= 192) {
+  if (M >= 256 && Nc >= 192 && K >= 128) {
     const int gm = (int)((M + 255) / 256), gn = (int)((Nc + 255) / 256);
-    igemm_kernel<256, 256, 64, 2, 4, AG, BG, OUT_BF16>
+    igemm_kernel_8ph<AG, BG, OUT_BF16>
         <<<gm * gn, 512, 0, s>>>(ag, bg, cptr, zp, M, Nc, K, gn);
   } else {
     const int gm = (int)((M + 127) / 128), gn = (int)((Nc + 127) / 128);
