@@ -88,14 +88,34 @@ def main() -> int:
         xs.append(x)
         ys.append(y.to(ctx.device))
 
-    def step(i: int):
+    def raw_step(x, y):
         engine.zero_grad()
-        logits = engine(xs[i % nbatch])
-        loss = F_amd.softmax_cross_entropy(logits, ys[i % nbatch])
+        logits = engine(x)
+        loss = F_amd.softmax_cross_entropy(logits, y)
         loss.backward()
         engine.finalize_backward()
         opt.step()
         return loss
+
+    holder = {"graphed": None}
+    if cuda:
+        from deeplearning_cfn_amd.engine.graph import GraphedStep, graphs_enabled
+        if graphs_enabled(n):
+            holder["graphed"] = GraphedStep(
+                lambda ins: raw_step(ins[0], ins[1]), [xs[0], ys[0]])
+
+    def step(i: int):
+        g = holder["graphed"]
+        if g is not None:
+            try:
+                return g.step([xs[i % nbatch], ys[i % nbatch]])
+            except Exception as e:  # capture unsupported -> eager fallback
+                import sys as _sys
+                print(f"[bench] hipGraph capture failed ({type(e).__name__}: "
+                      f"{e}); running eager", file=_sys.stderr)
+                holder["graphed"] = None
+                torch.cuda.synchronize()
+        return raw_step(xs[i % nbatch], ys[i % nbatch])
 
     def sync():
         if dist.is_initialized() and n > 1:
