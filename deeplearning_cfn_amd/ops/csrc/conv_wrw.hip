@@ -1,0 +1,246 @@
+// Implicit-GEMM NHWC convolution WEIGHT gradient (wrw) on MFMA.
+//
+// Completes the native conv triple (fwd/dgrad in conv.hip) — replaces the
+// library wgrad (SURVEY.md §3.2 "cuDNN conv kernels", backward-weights).
+//
+//   dW_cl[co, kf=(dh,dw,ci)] = sum_m  g[m, co] * x~[m, kf]
+//     m = (n, ho, wo);  x~[m, kf] = x[(n, ho*s-p+dh, wo*s-p+dw), ci]
+//
+// The MFMA reduction axis is m — but BOTH operands are m-major in memory
+// (channels-last: co / ci contiguous), so the LDS tiles are staged
+// m-row-major with global_load_lds (coalesced 16-B pieces) and the MFMA
+// fragments (which need per-lane m-contiguity) are read back with the
+// gfx950 hardware transpose read `ds_read_b64_tr_b16` (guide T10: the
+// attention-V recipe, [4m x 16c] subtile blocks, conflict-free groups).
+//
+// Split-M: each block owns a contiguous m-slice and accumulates its
+// fp32 128x128 output tile with global atomicAdd (unsafeAtomicAdd ->
+// global_atomic_add_f32); dW is zero-initialised per call.
+#include <torch/extension.h>
+#include <ATen/cuda/CUDAContext.h>
+
+#include "igemm_core.h"
+
+namespace dlx {
+
+namespace {
+
+using bf16v4 = __attribute__((ext_vector_type(4))) __bf16;
+using bf16x4 = __attribute__((ext_vector_type(4))) short;
+typedef __attribute__((address_space(3))) bf16v4 lds_bf16v4_t;
+
+// Transpose-read one MFMA k-half: lane l gets 4 consecutive m values of
+// its column (l&15) from subtile block (l>>4)*... — see image layout.
+DLX_DEV bf16x4 tr_read(const uint16_t* lds_base, int elem_off) {
+  auto* p = (lds_bf16v4_t*)(lds_base + elem_off);
+  bf16v4 v = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p);
+  return __builtin_bit_cast(bf16x4, v);
+}
+
+// LDS image per operand tile [BK m-rows x 128 cols], bf16:
+//   [BK/32 ksteps][8 cblocks][2 i][4 g][4 r][16 cc]   (64-elem blocks)
+// element (mloc, c):  kk=mloc>>5, m32=mloc&31, g=m32>>3, i=(m32&7)>>2,
+//                     r=m32&3, cb=c>>4, cc=c&15
+//   off = (((kk*8 + cb)*8) + i*4 + g)*64 + r*16 + cc
+// Within one (kk, cb) group the 4 lane-group blocks of read i are 64
+// elems (128 B = 32 dwords) apart -> groups 0/1 (and 2/3) hit disjoint
+// bank halves (guide §LDS: tr_b16 conflicts are per 32-lane half).
+constexpr int kBMw = 128;   // output rows (co) per block
+constexpr int kBNw = 128;   // output cols (kf) per block
+constexpr int kBKw = 64;    // m per K-tile
+constexpr int kThreadsW = 256;  // 4 waves, 2x2, 64x64 per wave
+constexpr int kSlotsW = kBKw * 128 / 8;  // 16-B slots per operand tile
+
+// decompose slot f -> (mloc, c) of the image above (all shifts)
+DLX_DEV void slot_to_mc(int f, int& mloc, int& c) {
+  const int e = f * 8;
+  const int blk = e >> 6;       // 64-elem block
+  const int rem = e & 63;       // r*16 + cc0
+  const int r = rem >> 4;
+  const int cc0 = rem & 15;     // 0 or 8
+  const int w = blk & 7;        // i*4 + g
+  const int i = w >> 2;
+  const int g = w & 3;
+  const int kkcb = blk >> 3;
+  const int cb = kkcb & 7;
+  const int kk = kkcb >> 3;
+  mloc = (kk << 5) + (g << 3) + (i << 2) + r;
+  c = (cb << 4) + cc0;
+}
+
+// element offset of the first element lane l reads for (kstep kk,
+// col-block cb, k-half i):  block (kk*8+cb)*8 + i*4 + (l>>4), col l&15
+DLX_DEV int frag_off(int kk, int cb, int i, int lane) {
+  return (((kk * 8 + cb) * 8) + i * 4 + (lane >> 4)) * 64 + (lane & 15);
+}
+
+template <class AG, class BG>
+__global__ __launch_bounds__(kThreadsW) void conv_wrw_kernel(
+    AG ag, BG bg, float* __restrict__ out,  // [BM-rows=co, BN-cols=kf]
+    const uint16_t* __restrict__ zero_page, long Mred, long Rows, long Cols,
+    int grid_n, int split, long m_per_slice) {
+  __shared__ uint16_t lds[2 * 2 * kSlotsW * 8];  // [buf][A|B][slots][8]
+
+  int nwg = gridDim.x;
+  int bid = blockIdx.x;
+  {  // XCD-aware bijective remap (guide T1)
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = bid % 8, idx = bid / 8;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  // tile-major: a tile's slices stay adjacent (same XCD chunk)
+  const int tile = bid / split, slice = bid % split;
+  const int bm = tile / grid_n, bn = tile % grid_n;
+  const long row0 = (long)bm * kBMw, col0 = (long)bn * kBNw;
+  const long mbeg = slice * m_per_slice;
+  const long mend = (mbeg + m_per_slice < Mred) ? mbeg + m_per_slice : Mred;
+  if (mbeg >= mend) return;
+
+  const int lane = threadIdx.x % kWave;
+  const int wid = threadIdx.x / kWave;
+  const int wr = wid >> 1, wc = wid & 1;  // 2x2 waves, 64x64 each
+  const int l15 = lane & 15, l4 = lane >> 4;
+
+  f32x4 acc[4][4] = {};
+  const long ntile = (mend - mbeg + kBKw - 1) / kBKw;
+
+  auto stage = [&](int buf, long t) {
+    const long mt0 = mbeg + t * kBKw;
+    uint16_t* la = lds + (buf * 2 + 0) * kSlotsW * 8;
+    uint16_t* lb = lds + (buf * 2 + 1) * kSlotsW * 8;
+#pragma unroll
+    for (int u = 0; u < kSlotsW / kThreadsW; ++u) {
+      const int f = u * kThreadsW + threadIdx.x;
+      int mloc, c;
+      slot_to_mc(f, mloc, c);
+      const long m = mt0 + mloc;
+      const uint16_t* sa =
+          (m < mend) ? ag.src(m, row0 + c, zero_page) : zero_page;
+      const uint16_t* sb =
+          (m < mend) ? bg.src(m, col0 + c, zero_page) : zero_page;
+      __builtin_amdgcn_global_load_lds(
+          (glb_u32_t*)sa, (lds_u32_t*)(la + (long)f * 8), 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (glb_u32_t*)sb, (lds_u32_t*)(lb + (long)f * 8), 16, 0, 0);
+    }
+  };
+
+  stage(0, 0);
+  __builtin_amdgcn_s_waitcnt(0);
+  __syncthreads();
+
+  int buf = 0;
+  for (long t = 0; t < ntile; ++t) {
+    if (t + 1 < ntile) stage(buf ^ 1, t + 1);
+    const uint16_t* la = lds + (buf * 2 + 0) * kSlotsW * 8;
+    const uint16_t* lb = lds + (buf * 2 + 1) * kSlotsW * 8;
+#pragma unroll
+    for (int kk = 0; kk < kBKw / 32; ++kk) {
+      // A fragments: rows (co) of this wave's 64-row strip
+      bf16x8 af[4], bfr[4];
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi) {
+        const int cb = wr * 4 + mi;
+        bf16x4 lo = tr_read(la, frag_off(kk, cb, 0, lane));
+        bf16x4 hi = tr_read(la, frag_off(kk, cb, 1, lane));
+        af[mi] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+      }
+#pragma unroll
+      for (int ni = 0; ni < 4; ++ni) {
+        const int cb = wc * 4 + ni;
+        bf16x4 lo = tr_read(lb, frag_off(kk, cb, 0, lane));
+        bf16x4 hi = tr_read(lb, frag_off(kk, cb, 1, lane));
+        bfr[ni] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mi = 0; mi < 4; ++mi)
+#pragma unroll
+        for (int ni = 0; ni < 4; ++ni)
+          acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    __syncthreads();
+    buf ^= 1;
+  }
+
+  // epilogue: D map col=lane&15, row=(lane>>4)*4+r; atomic accumulate
+#pragma unroll
+  for (int mi = 0; mi < 4; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < 4; ++ni) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long row = row0 + wr * 64 + mi * 16 + l4 * 4 + r;
+        const long col = col0 + wc * 64 + ni * 16 + l15;
+        if (row < Rows && col < Cols) {
+          if (split > 1)
+            unsafeAtomicAdd(&out[row * Cols + col], acc[mi][ni][r]);
+          else
+            out[row * Cols + col] = acc[mi][ni][r];
+        }
+      }
+    }
+  }
+}
+
+}  // namespace
+
+// g: [N,Co,Ho,Wo] CL bf16;  x: [N,Ci,H,W] CL bf16 ->
+// dW bf16 with channels_last strides of [Co,Ci,Kh,Kw]
+at::Tensor conv2d_wrw_impl(const at::Tensor& g_in, const at::Tensor& x_in,
+                           int64_t Kh, int64_t Kw, int64_t stride,
+                           int64_t pad) {
+  auto g = g_in.contiguous(at::MemoryFormat::ChannelsLast);
+  auto x = x_in.contiguous(at::MemoryFormat::ChannelsLast);
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16, "wrw: bf16 only");
+  ConvShape p;
+  p.N = x.size(0); p.Ci = x.size(1); p.H = x.size(2); p.W = x.size(3);
+  p.Co = g.size(1); p.Ho = g.size(2); p.Wo = g.size(3);
+  p.Kh = (int)Kh; p.Kw = (int)Kw;
+  p.stride = (int)stride; p.pad = (int)pad;
+  p.K = (long)p.Kh * p.Kw * p.Ci;   // fwd-gather k-extent (= kf here)
+  p.M = (long)p.N * p.Ho * p.Wo;    // reduction extent
+  TORCH_CHECK(p.Ci % 8 == 0, "wrw: Cin % 8 != 0 (use library path)");
+  TORCH_CHECK(p.Co % 8 == 0, "wrw: Cout % 8 != 0 (use library path)");
+
+  const long Rows = p.Co, Cols = p.K;
+  const int gm = (int)((Rows + kBMw - 1) / kBMw);
+  const int gn = (int)((Cols + kBNw - 1) / kBNw);
+  // split the m reduction so the grid reaches ~2 blocks/CU
+  long split = 768 / ((long)gm * gn);
+  const long max_split = (p.M + kBKw - 1) / kBKw;
+  if (split > max_split) split = max_split;
+  if (split < 1) split = 1;
+  long m_per_slice = (p.M + split - 1) / split;
+  // round slices to BK so every tile but the last is full
+  m_per_slice = ((m_per_slice + kBKw - 1) / kBKw) * kBKw;
+
+  auto outf = (split > 1)
+                  ? at::zeros({Rows, Cols}, x.options().dtype(at::kFloat))
+                  : at::empty({Rows, Cols}, x.options().dtype(at::kFloat));
+  static at::Tensor zp;  // cached 16-B zero page (per process)
+  if (!zp.defined() || zp.device() != x.device())
+    zp = at::zeros({16}, x.options().dtype(at::kBFloat16));
+  auto s = at::cuda::getCurrentCUDAStream();
+
+  // A-operand: g rows (m-major, co contiguous); B-operand: x~ im2col gather
+  RowGather ag{reinterpret_cast<const uint16_t*>(g.data_ptr()), p.M,
+               (long)p.Co};
+  ConvFwdGather bg{reinterpret_cast<const uint16_t*>(x.data_ptr()), p};
+  const int nblocks = gm * gn * (int)split;
+  conv_wrw_kernel<RowGather, ConvFwdGather>
+      <<<nblocks, kThreadsW, 0, s>>>(
+          ag, bg, outf.data_ptr<float>(),
+          reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, Rows, Cols,
+          gn, (int)split, m_per_slice);
+
+  // [Co, Kf] fp32 -> bf16 tensor with channels_last strides of the weight
+  return outf.to(at::kBFloat16)
+      .view({(long)p.Co, (long)p.Kh, (long)p.Kw, (long)p.Ci})
+      .permute({0, 3, 1, 2});
+}
+
+}  // namespace dlx

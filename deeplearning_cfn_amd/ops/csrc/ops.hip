@@ -69,6 +69,11 @@ at::Tensor conv2d_dgrad_impl(const at::Tensor& g, const at::Tensor& w,
                              int64_t H, int64_t W, int64_t stride,
                              int64_t pad);
 
+// conv_wrw.hip
+at::Tensor conv2d_wrw_impl(const at::Tensor& g, const at::Tensor& x,
+                           int64_t Kh, int64_t Kw, int64_t stride,
+                           int64_t pad);
+
 // layernorm.hip
 std::tuple<at::Tensor, at::Tensor, at::Tensor> ln_fwd_impl(
     const at::Tensor& x, const at::Tensor& w, const at::Tensor& b, double eps);
@@ -232,6 +237,7 @@ TORCH_LIBRARY(dlx_hip, m) {
   m.def("transpose2d(Tensor x) -> Tensor");
   m.def("conv2d_fwd(Tensor x, Tensor w, int stride, int pad) -> Tensor");
   m.def("conv2d_dgrad(Tensor g, Tensor w, int H, int W, int stride, int pad) -> Tensor");
+  m.def("conv2d_wrw(Tensor g, Tensor x, int Kh, int Kw, int stride, int pad) -> Tensor");
   m.def("ln_fwd(Tensor x, Tensor w, Tensor b, float eps) -> (Tensor, Tensor, Tensor)");
   m.def("ln_bwd(Tensor dy, Tensor x, Tensor w, Tensor mean, Tensor invstd) -> (Tensor, Tensor, Tensor)");
   m.def("gelu_fwd(Tensor x) -> Tensor");
@@ -263,6 +269,7 @@ TORCH_LIBRARY_IMPL(dlx_hip, CUDA, m) {
   m.impl("transpose2d", dlx::transpose2d_impl);
   m.impl("conv2d_fwd", dlx::conv2d_fwd_impl);
   m.impl("conv2d_dgrad", dlx::conv2d_dgrad_impl);
+  m.impl("conv2d_wrw", dlx::conv2d_wrw_impl);
   m.impl("ln_fwd", dlx::ln_fwd_impl);
   m.impl("ln_bwd", dlx::ln_bwd_impl);
   m.impl("gelu_fwd", dlx::gelu_fwd_impl);

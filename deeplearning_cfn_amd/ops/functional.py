@@ -183,10 +183,13 @@ class _Conv2d(torch.autograd.Function):
         if ctx.needs_input_grad[0]:
             dx = native().conv2d_dgrad(grad, w, x.shape[2], x.shape[3], s, p)
         if ctx.needs_input_grad[1]:
-            # wgrad stays on the library path until the rr-GEMM kernel lands
-            dw = torch.ops.aten.convolution_backward(
-                grad, x, w, None, [s, s], [p, p], [1, 1], False, [0, 0], 1,
-                [False, True, False])[1]
+            if _os.environ.get("DLX_CONV_WRW", "native") == "native":
+                # native MFMA wrw (m-reduction GEMM via ds_read_b64_tr_b16)
+                dw = native().conv2d_wrw(grad, x, w.shape[2], w.shape[3], s, p)
+            else:
+                dw = torch.ops.aten.convolution_backward(
+                    grad, x, w, None, [s, s], [p, p], [1, 1], False, [0, 0],
+                    1, [False, True, False])[1]
         return dx, dw, None, None
 
 

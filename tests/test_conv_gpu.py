@@ -57,6 +57,22 @@ def test_conv_dgrad_parity(shape):
                                atol=3e-2 * (co * k * k) ** 0.25)
 
 
+@pytest.mark.parametrize("shape", SHAPES)
+def test_conv_wrw_parity(shape):
+    n, ci, h, w, co, k, s, p = shape
+    x, wt = _mk(n, ci, h, w, co, k, seed=sum(shape) + 2)
+    wf = wt.float().requires_grad_()
+    ref = torch.nn.functional.conv2d(x.float(), wf, None, s, p)
+    g = torch.Generator(device="cuda").manual_seed(11)
+    dy = torch.randn(ref.shape, generator=g, device="cuda").bfloat16()
+    dy = dy.contiguous(memory_format=torch.channels_last)
+    ref.backward(dy.float())
+    dw = _native().conv2d_wrw(dy, x, k, k, s, p)
+    m = n * ref.shape[2] * ref.shape[3]
+    torch.testing.assert_close(dw.float(), wf.grad, rtol=4e-2,
+                               atol=2e-2 * m ** 0.5)
+
+
 def test_conv_autograd_function():
     from deeplearning_cfn_amd.ops import functional as F_amd
     n, ci, h, w, co, k, s, p = 2, 64, 14, 14, 128, 3, 1, 1
