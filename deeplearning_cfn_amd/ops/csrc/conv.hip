@@ -42,6 +42,7 @@ at::Tensor conv2d_fwd_impl(const at::Tensor& x_in, const at::Tensor& w_in,
   p.Wo = (p.W + 2 * p.pad - p.Kw) / p.stride + 1;
   p.K = (long)p.Kh * p.Kw * p.Ci;
   p.M = (long)p.N * p.Ho * p.Wo;
+  p.finalize();
   TORCH_CHECK(p.Ci % 8 == 0, "conv fwd: Cin % 8 != 0 (use library path)");
 
   auto y = at::empty({p.N, p.Co, p.Ho, p.Wo},
@@ -67,6 +68,7 @@ at::Tensor conv2d_dgrad_impl(const at::Tensor& g_in, const at::Tensor& w_in,
   p.H = (int)H; p.W = (int)W; p.stride = (int)stride; p.pad = (int)pad;
   p.K = (long)p.Kh * p.Kw * p.Co;
   p.M = (long)p.N * p.H * p.W;
+  p.finalize();
   TORCH_CHECK(p.Co % 8 == 0, "conv dgrad: Cout % 8 != 0");
 
   // W_t[ci, (dh,dw,co)] = W[co, ci, dh, dw]

@@ -68,10 +68,16 @@ DLX_DEV void slot_to_mc(int f, int& mloc, int& c) {
   c = (cb << 4) + cc0;
 }
 
-// element offset of the first element lane l reads for (kstep kk,
-// col-block cb, k-half i):  block (kk*8+cb)*8 + i*4 + (l>>4), col l&15
+// Per-lane element offset for the tr read of (kstep kk, col-block cb,
+// k-half i).  Probe-verified ds_read_b64_tr_b16 semantics (wave64):
+// within each 16-lane group, result lane l elem j = the 8-B read of lane
+// (4j + (l>>2)), element (l&3).  With per-lane address base + (l&15)*4
+// elems over a row-major [4m][16c] block, lane l therefore receives
+// column (l&15) across the 4 m-rows — the MFMA fragment gather.
+// Groups (l>>4) take consecutive blocks: lanes 0-31 cover 256 contiguous
+// bytes = all 64 LDS banks exactly once (conflict-free).
 DLX_DEV int frag_off(int kk, int cb, int i, int lane) {
-  return (((kk * 8 + cb) * 8) + i * 4 + (lane >> 4)) * 64 + (lane & 15);
+  return (((kk * 8 + cb) * 8) + i * 4 + (lane >> 4)) * 64 + (lane & 15) * 4;
 }
 
 template <class AG, class BG>
@@ -203,6 +209,7 @@ at::Tensor conv2d_wrw_impl(const at::Tensor& g_in, const at::Tensor& x_in,
   p.stride = (int)stride; p.pad = (int)pad;
   p.K = (long)p.Kh * p.Kw * p.Ci;   // fwd-gather k-extent (= kf here)
   p.M = (long)p.N * p.Ho * p.Wo;    // reduction extent
+  p.finalize();
   TORCH_CHECK(p.Ci % 8 == 0, "wrw: Cin % 8 != 0 (use library path)");
   TORCH_CHECK(p.Co % 8 == 0, "wrw: Cout % 8 != 0 (use library path)");
 

@@ -40,9 +40,39 @@ struct RowGather {
   }
 };
 
+// Division by a runtime-constant divisor via magic multiply (round-up
+// method: m = ceil(2^k/d), k = 31 + ceil_log2(d); exact for 0<=n<2^31).
+// A v_mul_hi_u64+shift replaces the ~20-cycle integer-divide expansion in
+// the per-slot gather address math (guide: VALU work beside glds staging).
+struct U32Div {
+  uint64_t mul;
+  int sh;
+  void init(uint32_t d) {
+    int k = 31;
+    while ((1ull << (k - 31)) < d) ++k;   // k = 31 + ceil_log2(d)
+    mul = ((1ull << k) + d - 1) / d;
+    sh = k;
+  }
+  DLX_DEV uint32_t div(uint32_t n) const {
+    return (uint32_t)(((uint64_t)n * mul) >> sh);
+  }
+};
+
 struct ConvShape {
   int N, H, W, Ci, Ho, Wo, Co, Kh, Kw, stride, pad;
   long K, M;
+  U32Div dv_wo, dv_hw, dv_ci, dv_kw;   // fwd/wrw gather divisors
+  U32Div dv_w, dv_hw_in, dv_co, dv_s;  // dgrad gather divisors
+  void finalize() {
+    dv_wo.init(Wo);
+    dv_hw.init(Ho * Wo);
+    dv_ci.init(Ci);
+    dv_kw.init(Kw);
+    dv_w.init(W);
+    dv_hw_in.init(H * W);
+    dv_co.init(Co);
+    dv_s.init(stride);
+  }
 };
 
 // conv fwd activations: row m=(n,ho,wo), k=(dh,dw,ci)
@@ -51,13 +81,14 @@ struct ConvFwdGather {
   ConvShape p;
   DLX_DEV const uint16_t* src(long m, long k, const uint16_t* zp) const {
     if (m >= p.M || k >= p.K) return zp;
-    const int hw = p.Ho * p.Wo;
-    const int n = (int)(m / hw);
-    const int rem = (int)(m - (long)n * hw);
-    const int ho = rem / p.Wo, wo = rem - (rem / p.Wo) * p.Wo;
-    const int dhw = (int)(k / p.Ci);
-    const int ci = (int)(k - (long)dhw * p.Ci);
-    const int dh = dhw / p.Kw, dw = dhw - (dhw / p.Kw) * p.Kw;
+    const int n = (int)p.dv_hw.div((uint32_t)m);
+    const int rem = (int)m - n * (p.Ho * p.Wo);
+    const int ho = (int)p.dv_wo.div((uint32_t)rem);
+    const int wo = rem - ho * p.Wo;
+    const int dhw = (int)p.dv_ci.div((uint32_t)k);
+    const int ci = (int)k - dhw * p.Ci;
+    const int dh = (int)p.dv_kw.div((uint32_t)dhw);
+    const int dw = dhw - dh * p.Kw;
     const int hi = ho * p.stride - p.pad + dh;
     const int wi = wo * p.stride - p.pad + dw;
     if (hi < 0 || hi >= p.H || wi < 0 || wi >= p.W) return zp;
@@ -71,17 +102,20 @@ struct ConvDgradGather {
   ConvShape p;
   DLX_DEV const uint16_t* src(long m, long k, const uint16_t* zp) const {
     if (m >= p.M || k >= p.K) return zp;
-    const int hw = p.H * p.W;
-    const int n = (int)(m / hw);
-    const int rem = (int)(m - (long)n * hw);
-    const int hi = rem / p.W, wi = rem - (rem / p.W) * p.W;
-    const int dhw = (int)(k / p.Co);
-    const int co = (int)(k - (long)dhw * p.Co);
-    const int dh = dhw / p.Kw, dw = dhw - (dhw / p.Kw) * p.Kw;
+    const int n = (int)p.dv_hw_in.div((uint32_t)m);
+    const int rem = (int)m - n * (p.H * p.W);
+    const int hi = (int)p.dv_w.div((uint32_t)rem);
+    const int wi = rem - hi * p.W;
+    const int dhw = (int)p.dv_co.div((uint32_t)k);
+    const int co = (int)k - dhw * p.Co;
+    const int dh = (int)p.dv_kw.div((uint32_t)dhw);
+    const int dw = dhw - dh * p.Kw;
     const int ht = hi + p.pad - dh;
     const int wt = wi + p.pad - dw;
-    if (ht < 0 || wt < 0 || (ht % p.stride) || (wt % p.stride)) return zp;
-    const int ho = ht / p.stride, wo = wt / p.stride;
+    if (ht < 0 || wt < 0) return zp;
+    const int ho = (int)p.dv_s.div((uint32_t)ht);
+    const int wo = (int)p.dv_s.div((uint32_t)wt);
+    if (ho * p.stride != ht || wo * p.stride != wt) return zp;
     if (ho >= p.Ho || wo >= p.Wo) return zp;
     return base + (((long)n * p.Ho + ho) * p.Wo + wo) * p.Co + co;
   }
