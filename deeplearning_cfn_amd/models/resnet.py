@@ -84,8 +84,9 @@ class BasicBlock(nn.Module):
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         identity = x if self.downsample is None else self.downsample(x)
         out = self.bn1(self.conv1(x))
-        out = self.bn2(self.conv2(out))
-        return F_amd.add_relu(out, identity)
+        out = self.conv2(out)
+        return F_amd.batch_norm_add_relu(out, self.bn2.bn, identity,
+                                         self.training)
 
 
 class Bottleneck(nn.Module):
@@ -106,8 +107,9 @@ class Bottleneck(nn.Module):
         identity = x if self.downsample is None else self.downsample(x)
         out = self.bn1(self.conv1(x))
         out = self.bn2(self.conv2(out))
-        out = self.bn3(self.conv3(out))
-        return F_amd.add_relu(out, identity)
+        out = self.conv3(out)
+        return F_amd.batch_norm_add_relu(out, self.bn3.bn, identity,
+                                         self.training)
 
 
 class ResNet(nn.Module):

@@ -166,11 +166,12 @@ __global__ void bn_eval_coeffs_kernel(const float* __restrict__ w,
 // ---- normalize: y = x*scale[c] + shift[c] (+ReLU) ----------------------
 // Same chunk/stripe geometry; scale/shift hoisted out of the row loop.
 
-template <typename T, bool RELU, bool VEC>
+template <typename T, bool RELU, bool VEC, bool RES = false>
 __global__ void bn_norm_kernel(const T* __restrict__ x, T* __restrict__ y,
                                const float* __restrict__ scale,
                                const float* __restrict__ shift, long R, int C,
-                               int CT, long rows_per_stripe) {
+                               int CT, long rows_per_stripe,
+                               const T* __restrict__ res = nullptr) {
   using VT = VecTraits<T>;
   using Vec = typename VT::Vec;
   constexpr int VW = VEC ? VT::width : 1;
@@ -190,15 +191,19 @@ __global__ void bn_norm_kernel(const T* __restrict__ x, T* __restrict__ y,
     const long i = r * C + c0;
     if (VEC) {
       Vec xv = *reinterpret_cast<const Vec*>(x + i);
+      Vec rv;
+      if (RES) rv = *reinterpret_cast<const Vec*>(res + i);
       Vec yv;
 #pragma unroll
       for (int j = 0; j < VW; ++j) {
         float v = VT::load(xv, j) * sc[j] + sh[j];
+        if (RES) v += VT::load(rv, j);
         VT::store(yv, j, RELU ? fmaxf(v, 0.f) : v);
       }
       *reinterpret_cast<Vec*>(y + i) = yv;
     } else {
       float v = VT::load1(x + i) * sc[0] + sh[0];
+      if (RES) v += VT::load1(res + i);
       VT::store1(y + i, RELU ? fmaxf(v, 0.f) : v);
     }
   }
@@ -309,13 +314,14 @@ __global__ void bn_finalize_bwd_kernel(const float* __restrict__ sums,
 
 // ---- dx pass: dx = cA*dy_eff + cB*x + cD (coefficients hoisted) --------
 
-template <typename T, bool RELU, bool VEC>
+template <typename T, bool RELU, bool VEC, bool DRES = false>
 __global__ void bn_dx_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                              const T* __restrict__ y,
                              const float* __restrict__ cA,
                              const float* __restrict__ cB,
                              const float* __restrict__ cD, T* __restrict__ dx,
-                             long R, int C, int CT, long rows_per_stripe) {
+                             long R, int C, int CT, long rows_per_stripe,
+                             T* __restrict__ dres = nullptr) {
   using VT = VecTraits<T>;
   using Vec = typename VT::Vec;
   constexpr int VW = VEC ? VT::width : 1;
@@ -339,17 +345,20 @@ __global__ void bn_dx_kernel(const T* __restrict__ dy, const T* __restrict__ x,
       Vec xv = *reinterpret_cast<const Vec*>(x + i);
       Vec yv;
       if (RELU) yv = *reinterpret_cast<const Vec*>(y + i);
-      Vec ov;
+      Vec ov, rv;
 #pragma unroll
       for (int j = 0; j < VW; ++j) {
         float g = VT::load(gv, j);
         if (RELU && VT::load(yv, j) <= 0.f) g = 0.f;
+        if (DRES) VT::store(rv, j, g);
         VT::store(ov, j, A[j] * g + B[j] * VT::load(xv, j) + D[j]);
       }
       *reinterpret_cast<Vec*>(dx + i) = ov;
+      if (DRES) *reinterpret_cast<Vec*>(dres + i) = rv;
     } else {
       float g = VT::load1(dy + i);
       if (RELU && VT::load1(y + i) <= 0.f) g = 0.f;
+      if (DRES) VT::store1(dres + i, g);
       VT::store1(dx + i, A[0] * g + B[0] * VT::load1(x + i) + D[0]);
     }
   }
@@ -497,6 +506,124 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_bwd_impl(
   else { DLX_BN_BWD(float, false, false) }
 #undef DLX_BN_BWD
   return {dx, dgamma, dbeta};
+}
+
+// ---- fused BN + residual add + ReLU (ResNet block tail) -----------------
+// fwd: y = relu(x*scale + shift + residual); bwd additionally returns
+// dres = relu-masked dy in the same dx pass (one fewer elementwise pass
+// and two fewer kernels than bn + add_relu).
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd_train_res_impl(
+    const at::Tensor& x_in, const at::Tensor& w, const at::Tensor& b,
+    at::Tensor& rmean, at::Tensor& rvar, double momentum, double eps,
+    const at::Tensor& res_in) {
+  auto x = x_in.contiguous(at::MemoryFormat::ChannelsLast);
+  auto res = res_in.contiguous(at::MemoryFormat::ChannelsLast);
+  auto s = at::cuda::getCurrentCUDAStream();
+  const bool bf = x.scalar_type() == at::kBFloat16;
+  auto sh = bn_shape(x, bf ? 8 : 4);
+  TORCH_CHECK(sh.vec, "bn_res: C must be a multiple of the vector width");
+  auto fopts = x.options().dtype(at::kFloat);
+  auto sums = at::empty({(long)sh.stripes * 2 * sh.C}, fopts);
+  auto save_mean = at::empty({sh.C}, fopts);
+  auto save_invstd = at::empty({sh.C}, fopts);
+  auto scale = at::empty({sh.C}, fopts);
+  auto shift = at::empty({sh.C}, fopts);
+  auto y = at::empty_like(x);
+
+#define DLX_BN_RES(T)                                                        \
+  {                                                                          \
+    bn_reduce_fwd_kernel<T, true><<<sh.grid, kBlock, sh.lds, s>>>(           \
+        tp<T>(x), sums.data_ptr<float>(), sh.R, sh.C, sh.CT,                 \
+        sh.rows_per_stripe);                                                 \
+    bn_finalize_fwd_kernel<<<sh.C, kBlock, 0, s>>>(                          \
+        sums.data_ptr<float>(), w.data_ptr<float>(), b.data_ptr<float>(),    \
+        rmean.data_ptr<float>(), rvar.data_ptr<float>(),                     \
+        save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),          \
+        scale.data_ptr<float>(), shift.data_ptr<float>(), sh.R, sh.C,        \
+        sh.stripes, (float)momentum, (float)eps);                            \
+    bn_norm_kernel<T, true, true, true><<<sh.grid, kBlock, 0, s>>>(          \
+        tp<T>(x), tpm<T>(y), scale.data_ptr<float>(),                        \
+        shift.data_ptr<float>(), sh.R, sh.C, sh.CT, sh.rows_per_stripe,     \
+        tp<T>(res));                                                         \
+  }
+  if (bf) { DLX_BN_RES(uint16_t) } else { DLX_BN_RES(float) }
+#undef DLX_BN_RES
+  return {y, save_mean, save_invstd};
+}
+
+at::Tensor bn_fwd_eval_res_impl(const at::Tensor& x_in, const at::Tensor& w,
+                                const at::Tensor& b, const at::Tensor& rmean,
+                                const at::Tensor& rvar, double eps,
+                                const at::Tensor& res_in) {
+  auto x = x_in.contiguous(at::MemoryFormat::ChannelsLast);
+  auto res = res_in.contiguous(at::MemoryFormat::ChannelsLast);
+  auto s = at::cuda::getCurrentCUDAStream();
+  const bool bf = x.scalar_type() == at::kBFloat16;
+  auto sh = bn_shape(x, bf ? 8 : 4);
+  TORCH_CHECK(sh.vec, "bn_res: C must be a multiple of the vector width");
+  auto fopts = x.options().dtype(at::kFloat);
+  auto scale = at::empty({sh.C}, fopts);
+  auto shift = at::empty({sh.C}, fopts);
+  bn_eval_coeffs_kernel<<<(sh.C + 255) / 256, 256, 0, s>>>(
+      w.data_ptr<float>(), b.data_ptr<float>(), rmean.data_ptr<float>(),
+      rvar.data_ptr<float>(), scale.data_ptr<float>(), shift.data_ptr<float>(),
+      sh.C, (float)eps);
+  auto y = at::empty_like(x);
+  if (bf)
+    bn_norm_kernel<uint16_t, true, true, true><<<sh.grid, kBlock, 0, s>>>(
+        tp<uint16_t>(x), tpm<uint16_t>(y), scale.data_ptr<float>(),
+        shift.data_ptr<float>(), sh.R, sh.C, sh.CT, sh.rows_per_stripe,
+        tp<uint16_t>(res));
+  else
+    bn_norm_kernel<float, true, true, true><<<sh.grid, kBlock, 0, s>>>(
+        tp<float>(x), tpm<float>(y), scale.data_ptr<float>(),
+        shift.data_ptr<float>(), sh.R, sh.C, sh.CT, sh.rows_per_stripe,
+        tp<float>(res));
+  return y;
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_bwd_res_impl(
+    const at::Tensor& dy_in, const at::Tensor& x_in, const at::Tensor& w,
+    const at::Tensor& save_mean, const at::Tensor& save_invstd,
+    const at::Tensor& y_in) {
+  auto x = x_in.contiguous(at::MemoryFormat::ChannelsLast);
+  auto dy = dy_in.contiguous(at::MemoryFormat::ChannelsLast);
+  auto y = y_in.contiguous(at::MemoryFormat::ChannelsLast);
+  auto s = at::cuda::getCurrentCUDAStream();
+  const bool bf = x.scalar_type() == at::kBFloat16;
+  auto sh = bn_shape(x, bf ? 8 : 4);
+  TORCH_CHECK(sh.vec, "bn_res: C must be a multiple of the vector width");
+  auto fopts = x.options().dtype(at::kFloat);
+  auto sums = at::empty({(long)sh.stripes * 2 * sh.C}, fopts);
+  auto dgamma = at::empty({sh.C}, fopts);
+  auto dbeta = at::empty({sh.C}, fopts);
+  auto cA = at::empty({sh.C}, fopts);
+  auto cB = at::empty({sh.C}, fopts);
+  auto cD = at::empty({sh.C}, fopts);
+  auto dx = at::empty_like(x);
+  auto dres = at::empty_like(x);
+
+#define DLX_BN_BWDR(T)                                                       \
+  {                                                                          \
+    bn_reduce_bwd_kernel<T, true, true><<<sh.grid, kBlock, sh.lds, s>>>(     \
+        tp<T>(dy), tp<T>(x), tp<T>(y), save_mean.data_ptr<float>(),          \
+        save_invstd.data_ptr<float>(), sums.data_ptr<float>(), sh.R, sh.C,   \
+        sh.CT, sh.rows_per_stripe);                                          \
+    bn_finalize_bwd_kernel<<<sh.C, kBlock, 0, s>>>(                          \
+        sums.data_ptr<float>(), w.data_ptr<float>(),                         \
+        save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),          \
+        dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),                   \
+        cA.data_ptr<float>(), cB.data_ptr<float>(), cD.data_ptr<float>(),    \
+        sh.R, sh.C, sh.stripes);                                             \
+    bn_dx_kernel<T, true, true, true><<<sh.grid, kBlock, 0, s>>>(            \
+        tp<T>(dy), tp<T>(x), tp<T>(y), cA.data_ptr<float>(),                 \
+        cB.data_ptr<float>(), cD.data_ptr<float>(), tpm<T>(dx), sh.R, sh.C,  \
+        sh.CT, sh.rows_per_stripe, tpm<T>(dres));                            \
+  }
+  if (bf) { DLX_BN_BWDR(uint16_t) } else { DLX_BN_BWDR(float) }
+#undef DLX_BN_BWDR
+  return {dx, dgamma, dbeta, dres};
 }
 
 }  // namespace dlx

@@ -45,6 +45,18 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd_train_impl(
 at::Tensor bn_fwd_eval_impl(const at::Tensor& x, const at::Tensor& w,
                             const at::Tensor& b, const at::Tensor& rmean,
                             const at::Tensor& rvar, double eps, bool relu);
+std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd_train_res_impl(
+    const at::Tensor& x, const at::Tensor& w, const at::Tensor& b,
+    at::Tensor& rmean, at::Tensor& rvar, double momentum, double eps,
+    const at::Tensor& res);
+at::Tensor bn_fwd_eval_res_impl(const at::Tensor& x, const at::Tensor& w,
+                                const at::Tensor& b, const at::Tensor& rmean,
+                                const at::Tensor& rvar, double eps,
+                                const at::Tensor& res);
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_bwd_res_impl(
+    const at::Tensor& dy, const at::Tensor& x, const at::Tensor& w,
+    const at::Tensor& save_mean, const at::Tensor& save_invstd,
+    const at::Tensor& y);
 std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_bwd_impl(
     const at::Tensor& dy, const at::Tensor& x, const at::Tensor& w,
     const at::Tensor& save_mean, const at::Tensor& save_invstd,
@@ -225,6 +237,9 @@ TORCH_LIBRARY(dlx_hip, m) {
   m.def("bn_fwd_train(Tensor x, Tensor w, Tensor b, Tensor(a!) rmean, Tensor(b!) rvar, float momentum, float eps, bool relu) -> (Tensor, Tensor, Tensor)");
   m.def("bn_fwd_eval(Tensor x, Tensor w, Tensor b, Tensor rmean, Tensor rvar, float eps, bool relu) -> Tensor");
   m.def("bn_bwd(Tensor dy, Tensor x, Tensor w, Tensor save_mean, Tensor save_invstd, Tensor y, bool relu) -> (Tensor, Tensor, Tensor)");
+  m.def("bn_fwd_train_res(Tensor x, Tensor w, Tensor b, Tensor(a!) rmean, Tensor(b!) rvar, float momentum, float eps, Tensor res) -> (Tensor, Tensor, Tensor)");
+  m.def("bn_fwd_eval_res(Tensor x, Tensor w, Tensor b, Tensor rmean, Tensor rvar, float eps, Tensor res) -> Tensor");
+  m.def("bn_bwd_res(Tensor dy, Tensor x, Tensor w, Tensor save_mean, Tensor save_invstd, Tensor y) -> (Tensor, Tensor, Tensor, Tensor)");
   m.def("gemm_nt(Tensor a, Tensor b) -> Tensor");
   m.def("gemm_nn(Tensor g, Tensor w) -> Tensor");
   m.def("gemm_tn(Tensor g, Tensor x) -> Tensor");
@@ -257,6 +272,9 @@ TORCH_LIBRARY_IMPL(dlx_hip, CUDA, m) {
   m.impl("bn_fwd_train", dlx::bn_fwd_train_impl);
   m.impl("bn_fwd_eval", dlx::bn_fwd_eval_impl);
   m.impl("bn_bwd", dlx::bn_bwd_impl);
+  m.impl("bn_fwd_train_res", dlx::bn_fwd_train_res_impl);
+  m.impl("bn_fwd_eval_res", dlx::bn_fwd_eval_res_impl);
+  m.impl("bn_bwd_res", dlx::bn_bwd_res_impl);
   m.impl("gemm_nt", dlx::gemm_nt);
   m.impl("gemm_nn", dlx::gemm_nn);
   m.impl("gemm_tn", dlx::gemm_tn);

@@ -114,6 +114,52 @@ def batch_norm_act(x: torch.Tensor, bn: torch.nn.BatchNorm2d,
 
 
 # ---------------------------------------------------------------------------
+# fused BatchNorm + residual add + ReLU (ResNet block tail)
+# ---------------------------------------------------------------------------
+
+class _BNAddReLU(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, residual, weight, bias, running_mean, running_var,
+                momentum: float, eps: float):
+        x = _to_nhwc(x)
+        y, save_mean, save_invstd = native().bn_fwd_train_res(
+            x, weight, bias, running_mean, running_var, momentum, eps,
+            _to_nhwc(residual))
+        ctx.save_for_backward(x, weight, save_mean, save_invstd, y)
+        return y
+
+    @staticmethod
+    def backward(ctx, grad: torch.Tensor):
+        x, weight, save_mean, save_invstd, y = ctx.saved_tensors
+        dx, dweight, dbias, dres = native().bn_bwd_res(
+            _to_nhwc(grad), x, weight, save_mean, save_invstd, y)
+        return dx, dres, dweight, dbias, None, None, None, None
+
+
+def batch_norm_add_relu(x: torch.Tensor, bn: torch.nn.BatchNorm2d,
+                        residual: torch.Tensor,
+                        training: bool = True) -> torch.Tensor:
+    """relu(batch_norm(x) + residual) — the ResNet block tail, fused.
+
+    Native path: BN stats reduce + one normalize/add/relu pass; backward
+    emits dx AND the relu-masked residual grad in a single pass
+    (replaces cuDNN BN + framework add + ReLU).
+    """
+    if _use_native(x) and x.shape[1] % (8 if x.dtype == torch.bfloat16
+                                        else 4) == 0:
+        if training:
+            return _BNAddReLU.apply(x, residual, bn.weight, bn.bias,
+                                    bn.running_mean, bn.running_var,
+                                    bn.momentum, bn.eps)
+        return native().bn_fwd_eval_res(_to_nhwc(x), bn.weight, bn.bias,
+                                        bn.running_mean, bn.running_var,
+                                        bn.eps, _to_nhwc(residual))
+    out = F.batch_norm(x, bn.running_mean, bn.running_var, bn.weight,
+                       bn.bias, training, bn.momentum, bn.eps)
+    return F.relu(out + residual)
+
+
+# ---------------------------------------------------------------------------
 # fused softmax + cross-entropy (the classifier head)
 # ---------------------------------------------------------------------------
 

@@ -180,3 +180,41 @@ def test_adamw_flat_matches_torch():
         bc1, bc2 = 1 - 0.9 ** t, 1 - 0.999 ** t
         n.adamw_flat(p, g, m, v, 1e-3, 0.9, 0.999, 1e-8, 0.01, bc1, bc2)
     torch.testing.assert_close(p, p_ref.detach(), rtol=1e-5, atol=1e-6)
+
+
+def test_bn_add_relu_fused():
+    """Fused BN+residual+ReLU (fwd+bwd) vs plain fp32 torch composition."""
+    from deeplearning_cfn_amd.ops import functional as F_amd
+    torch.manual_seed(5)
+    N, C, H, W = 4, 64, 14, 14
+    x = torch.randn(N, C, H, W, device="cuda").bfloat16()
+    x = x.contiguous(memory_format=torch.channels_last).requires_grad_()
+    res = torch.randn(N, C, H, W, device="cuda").bfloat16()
+    res = res.contiguous(memory_format=torch.channels_last).requires_grad_()
+    bn = torch.nn.BatchNorm2d(C).cuda()
+    with torch.no_grad():
+        bn.weight.uniform_(0.5, 1.5)
+        bn.bias.uniform_(-0.5, 0.5)
+    y = F_amd.batch_norm_add_relu(x, bn, res, training=True)
+
+    xf = x.detach().float().requires_grad_()
+    rf = res.detach().float().requires_grad_()
+    bnf = torch.nn.BatchNorm2d(C).cuda()
+    with torch.no_grad():
+        bnf.weight.copy_(bn.weight)
+        bnf.bias.copy_(bn.bias)
+    ref = torch.relu(torch.nn.functional.batch_norm(
+        xf, bnf.running_mean, bnf.running_var, bnf.weight, bnf.bias,
+        True, 0.1, 1e-5) + rf)
+    torch.testing.assert_close(y.float(), ref, rtol=3e-2, atol=3e-2)
+
+    g = torch.randn_like(ref)
+    y.backward(g.bfloat16())
+    ref.backward(g)
+    torch.testing.assert_close(x.grad.float(), xf.grad, rtol=5e-2, atol=5e-2)
+    torch.testing.assert_close(res.grad.float(), rf.grad, rtol=5e-2,
+                               atol=5e-2)
+    torch.testing.assert_close(bn.weight.grad, bnf.weight.grad, rtol=2e-2,
+                               atol=2e-1)
+    torch.testing.assert_close(bn.bias.grad, bnf.bias.grad, rtol=2e-2,
+                               atol=2e-1)
