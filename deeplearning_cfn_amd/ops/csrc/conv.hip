@@ -74,10 +74,66 @@ at::Tensor conv2d_dgrad_impl(const at::Tensor& g_in, const at::Tensor& w_in,
   // W_t[ci, (dh,dw,co)] = W[co, ci, dh, dw]
   auto wt = w_in.permute({1, 2, 3, 0}).contiguous()
                  .view({p.Ci, (long)p.Kh * p.Kw * p.Co});
-  auto dx = at::empty({p.N, p.Ci, p.H, p.W},
-                      g.options().memory_format(at::MemoryFormat::ChannelsLast));
   auto zp = zero_page_for(g);
   auto s = at::cuda::getCurrentCUDAStream();
+
+  if (p.stride == 2) {
+    // Parity decomposition: 4 classes, each a dense GEMM over only the
+    // taps that can reach its pixels — 4x fewer MACs than zero-padding.
+    // A class with no taps (e.g. 1x1/s2: three of four) leaves dx zero,
+    // so pre-zero only in that case.
+    bool any_empty = false;
+    for (int a = 0; a < 2; ++a)
+      for (int b = 0; b < 2; ++b)
+        if (p.Kh <= ((a + p.pad) & 1) || p.Kw <= ((b + p.pad) & 1))
+          any_empty = true;
+    auto dx = any_empty
+                  ? at::zeros({p.N, p.Ci, p.H, p.W},
+                              g.options().memory_format(
+                                  at::MemoryFormat::ChannelsLast))
+                  : at::empty({p.N, p.Ci, p.H, p.W},
+                              g.options().memory_format(
+                                  at::MemoryFormat::ChannelsLast));
+    auto wt4 = wt.view({(long)p.Ci, (long)p.Kh, (long)p.Kw, (long)p.Co});
+    for (int a = 0; a < 2; ++a) {
+      for (int b = 0; b < 2; ++b) {
+        const int pa = (a + p.pad) & 1, pb = (b + p.pad) & 1;
+        const int nKh = (p.Kh > pa) ? (p.Kh - pa + 1) / 2 : 0;
+        const int nKw = (p.Kw > pb) ? (p.Kw - pb + 1) / 2 : 0;
+        if (nKh == 0 || nKw == 0) continue;  // dx stays zero here
+        const int Ha = (p.H > a) ? (p.H - a + 1) / 2 : 0;
+        const int Wb = (p.W > b) ? (p.W - b + 1) / 2 : 0;
+        if (Ha == 0 || Wb == 0) continue;
+        using at::indexing::Slice;
+        auto wt_ab = wt4.index({Slice(), Slice(pa, p.Kh, 2),
+                                Slice(pb, p.Kw, 2), Slice()})
+                         .contiguous()
+                         .view({(long)p.Ci, (long)nKh * nKw * p.Co});
+        ConvDgradS2Gather ag;
+        ag.base = reinterpret_cast<const uint16_t*>(g.data_ptr());
+        ag.p = p;
+        ag.a = a; ag.b = b; ag.pa = pa; ag.pb = pb; ag.nKw = nKw;
+        ag.Wb = Wb; ag.HaWb = Ha * Wb;
+        ag.dv_wb.init(Wb); ag.dv_hwb.init(Ha * Wb); ag.dv_nkw.init(nKw);
+        ag.M2 = (long)p.N * Ha * Wb;
+        ag.K2 = (long)nKh * nKw * p.Co;
+        RowGather bg{reinterpret_cast<const uint16_t*>(wt_ab.data_ptr()),
+                     p.Ci, ag.K2};
+        Stride2Out om;
+        om.H = p.H; om.W = p.W; om.Ci = p.Ci; om.a = a; om.b = b;
+        om.Wb = Wb; om.HaWb = Ha * Wb;
+        om.dv_wb.init(Wb); om.dv_hwb.init(Ha * Wb);
+        igemm_launch_om<ConvDgradS2Gather, RowGather, true, Stride2Out>(
+            ag, bg, dx.data_ptr(),
+            reinterpret_cast<const uint16_t*>(zp.data_ptr()), ag.M2, p.Ci,
+            ag.K2, s, om);
+      }
+    }
+    return dx;
+  }
+
+  auto dx = at::empty({p.N, p.Ci, p.H, p.W},
+                      g.options().memory_format(at::MemoryFormat::ChannelsLast));
   ConvDgradGather ag{reinterpret_cast<const uint16_t*>(g.data_ptr()), p};
   RowGather bg{reinterpret_cast<const uint16_t*>(wt.data_ptr()), p.Ci, p.K};
   igemm_launch<ConvDgradGather, RowGather, true>(

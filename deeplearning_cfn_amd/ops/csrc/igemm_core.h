@@ -121,14 +121,67 @@ struct ConvDgradGather {
   }
 };
 
+// stride-2 dgrad, parity-decomposed: class (a,b) covers input pixels
+// (2i+a, 2j+b); only taps with dh = pa+2*di, dw = pb+2*dj (pa=(a+pad)&1)
+// can hit them, so k enumerates (di, dj, co) — 4x fewer MACs than the
+// generic gather's zero-page-padded taps.  The B operand is the matching
+// weight slice wt[Ci, nKh, nKw, Co] (host-sliced, RowGather).
+struct ConvDgradS2Gather {
+  const uint16_t* base;  // g [N,Ho,Wo,Co] channels-last
+  ConvShape p;
+  int a, b, pa, pb, nKw, Wb, HaWb;
+  U32Div dv_wb, dv_hwb, dv_nkw;
+  long M2, K2;
+  DLX_DEV const uint16_t* src(long m, long k, const uint16_t* zp) const {
+    if (m >= M2 || k >= K2) return zp;
+    const int n = (int)dv_hwb.div((uint32_t)m);
+    const int rem = (int)m - n * HaWb;
+    const int i = (int)dv_wb.div((uint32_t)rem);
+    const int j = rem - i * Wb;
+    const int dij = (int)p.dv_co.div((uint32_t)k);
+    const int co = (int)k - dij * p.Co;
+    const int di = (int)dv_nkw.div((uint32_t)dij);
+    const int dj = dij - di * nKw;
+    const int ht = (2 * i + a) + p.pad - (pa + 2 * di);  // even
+    const int wt = (2 * j + b) + p.pad - (pb + 2 * dj);
+    if (ht < 0 || wt < 0) return zp;
+    const int ho = ht >> 1, wo = wt >> 1;
+    if (ho >= p.Ho || wo >= p.Wo) return zp;
+    return base + (((long)n * p.Ho + ho) * p.Wo + wo) * p.Co + co;
+  }
+};
+
+// ---- output row mappers ------------------------------------------------
+// Map an output row index to its element offset in C (row stride Nc for
+// the plain case; a scattered pixel address for the stride-2 dgrad
+// parity classes, where row enumerates only pixels (2i+a, 2j+b)).
+
+struct IdentityOut {
+  long Nc;
+  DLX_DEV long off(long row) const { return row * Nc; }
+};
+
+struct Stride2Out {
+  int H, W, Ci, a, b, Wb;   // Wb = pixels per row in this class
+  U32Div dv_wb, dv_hwb;     // /Wb, /(Ha*Wb)
+  int HaWb;
+  DLX_DEV long off(long row) const {
+    const int n = (int)dv_hwb.div((uint32_t)row);
+    const int rem = (int)row - n * HaWb;
+    const int i = (int)dv_wb.div((uint32_t)rem);
+    const int j = rem - i * Wb;
+    return (((long)n * H + 2 * i + a) * W + 2 * j + b) * Ci;
+  }
+};
+
 // ---- the kernel --------------------------------------------------------
 
 template <int BM, int BN, int BK, int WM, int WN,
-          class AG, class BG, bool OUT_BF16>
+          class AG, class BG, bool OUT_BF16, class OM = IdentityOut>
 __global__ __launch_bounds__(WM* WN * 64) void igemm_kernel(
     AG ag, BG bg, void* __restrict__ Cout_,
     const uint16_t* __restrict__ zero_page,
-    long M, long Nc, long K, int grid_n) {
+    long M, long Nc, long K, int grid_n, OM om) {
   constexpr int THREADS = WM * WN * 64;
   constexpr int SLOTS_A = (BK / 8) * BM;
   constexpr int SLOTS_B = (BK / 8) * BN;
@@ -230,11 +283,11 @@ __global__ __launch_bounds__(WM* WN * 64) void igemm_kernel(
         long row = m0 + wr * (BM / WM) + mi * 16 + l4 * 4 + r;
         long col = n0 + wc * (BN / WN) + ni * 16 + l15;
         if (row < M && col < Nc) {
+          const long o = om.off(row) + col;
           if (OUT_BF16)
-            reinterpret_cast<uint16_t*>(Cout_)[row * Nc + col] =
-                f2bf(acc[mi][ni][r]);
+            reinterpret_cast<uint16_t*>(Cout_)[o] = f2bf(acc[mi][ni][r]);
           else
-            reinterpret_cast<float*>(Cout_)[row * Nc + col] = acc[mi][ni][r];
+            reinterpret_cast<float*>(Cout_)[o] = acc[mi][ni][r];
         }
       }
     }
@@ -255,11 +308,11 @@ __global__ __launch_bounds__(WM* WN * 64) void igemm_kernel(
 //    creates the wave role diversity it needs).
 // ---------------------------------------------------------------------
 
-template <class AG, class BG, bool OUT_BF16>
+template <class AG, class BG, bool OUT_BF16, class OM = IdentityOut>
 __global__ __launch_bounds__(512) void igemm_kernel_8ph(
     AG ag, BG bg, void* __restrict__ Cout_,
     const uint16_t* __restrict__ zero_page,
-    long M, long Nc, long K, int grid_n) {
+    long M, long Nc, long K, int grid_n, OM om) {
   constexpr int BM = 256, BN = 256, BK = 64;
   constexpr int THREADS = 512;
   constexpr int SLOTS = (BK / 8) * BM;          // 2048 16-B slots/operand
@@ -381,11 +434,11 @@ __global__ __launch_bounds__(512) void igemm_kernel_8ph(
         long row = m0 + wr * 128 + mi * 16 + l4 * 4 + r;
         long col = n0 + wc * 64 + ni * 16 + l15;
         if (row < M && col < Nc) {
+          const long o = om.off(row) + col;
           if (OUT_BF16)
-            reinterpret_cast<uint16_t*>(Cout_)[row * Nc + col] =
-                f2bf(acc[mi][ni][r]);
+            reinterpret_cast<uint16_t*>(Cout_)[o] = f2bf(acc[mi][ni][r]);
           else
-            reinterpret_cast<float*>(Cout_)[row * Nc + col] = acc[mi][ni][r];
+            reinterpret_cast<float*>(Cout_)[o] = acc[mi][ni][r];
         }
       }
     }
@@ -394,19 +447,27 @@ __global__ __launch_bounds__(512) void igemm_kernel_8ph(
 
 // launch helper: picks the 8-phase 256^2 pipeline for big problems,
 // the simple 128^2 kernel otherwise
+template <class AG, class BG, bool OUT_BF16, class OM>
+inline void igemm_launch_om(const AG& ag, const BG& bg, void* cptr,
+                            const uint16_t* zp, long M, long Nc, long K,
+                            hipStream_t s, const OM& om) {
+  if (M >= 256 && Nc >= 192 && K >= 128) {
+    const int gm = (int)((M + 255) / 256), gn = (int)((Nc + 255) / 256);
+    igemm_kernel_8ph<AG, BG, OUT_BF16, OM>
+        <<<gm * gn, 512, 0, s>>>(ag, bg, cptr, zp, M, Nc, K, gn, om);
+  } else {
+    const int gm = (int)((M + 127) / 128), gn = (int)((Nc + 127) / 128);
+    igemm_kernel<128, 128, 64, 2, 2, AG, BG, OUT_BF16, OM>
+        <<<gm * gn, 256, 0, s>>>(ag, bg, cptr, zp, M, Nc, K, gn, om);
+  }
+}
+
 template <class AG, class BG, bool OUT_BF16>
 inline void igemm_launch(const AG& ag, const BG& bg, void* cptr,
                          const uint16_t* zp, long M, long Nc, long K,
                          hipStream_t s) {
-  if (M >= 256 && Nc >= 192 && K >= 128) {
-    const int gm = (int)((M + 255) / 256), gn = (int)((Nc + 255) / 256);
-    igemm_kernel_8ph<AG, BG, OUT_BF16>
-        <<<gm * gn, 512, 0, s>>>(ag, bg, cptr, zp, M, Nc, K, gn);
-  } else {
-    const int gm = (int)((M + 127) / 128), gn = (int)((Nc + 127) / 128);
-    igemm_kernel<128, 128, 64, 2, 2, AG, BG, OUT_BF16>
-        <<<gm * gn, 256, 0, s>>>(ag, bg, cptr, zp, M, Nc, K, gn);
-  }
+  igemm_launch_om<AG, BG, OUT_BF16, IdentityOut>(ag, bg, cptr, zp, M, Nc, K,
+                                                 s, IdentityOut{Nc});
 }
 
 }  // namespace dlx

@@ -16,7 +16,10 @@ namespace dlx {
 
 namespace {
 
-struct alignas(8) U8x8 { uint8_t v[8]; };
+template <int W> struct TapVec;
+template <> struct alignas(8) TapVec<8> { uint8_t v[8]; };
+template <> struct alignas(4) TapVec<4> { uint8_t v[4]; };
+template <> struct alignas(1) TapVec<1> { uint8_t v[1]; };
 
 // ---- maxpool 3x3 s2 p1 fwd: y + per-element tap byte -------------------
 
@@ -78,14 +81,14 @@ __global__ void maxpool_fwd_kernel(const T* __restrict__ x, T* __restrict__ y,
     const long o = (((long)nb * Ho + ho) * Wo + wo) * C + c0;
     if (VEC) {
       Vec ov;
-      U8x8 tv;
+      TapVec<VW> tv;
 #pragma unroll
       for (int j = 0; j < VW; ++j) {
         VT::store(ov, j, best[j]);
         tv.v[j] = tap[j];
       }
       *reinterpret_cast<Vec*>(y + o) = ov;
-      *reinterpret_cast<U8x8*>(argtap + o) = tv;
+      *reinterpret_cast<TapVec<VW>*>(argtap + o) = tv;
     } else {
       VT::store1(y + o, best[0]);
       argtap[o] = tap[0];
@@ -128,7 +131,7 @@ __global__ void maxpool_bwd_kernel(const T* __restrict__ dy,
         const uint8_t want = (uint8_t)(dh * 3 + (wi - (wo * 2 - 1)));
         const long oi = (((long)nb * Ho + ho) * Wo + wo) * C + c0;
         if (VEC) {
-          U8x8 tv = *reinterpret_cast<const U8x8*>(argtap + oi);
+          TapVec<VW> tv = *reinterpret_cast<const TapVec<VW>*>(argtap + oi);
           Vec gv = *reinterpret_cast<const Vec*>(dy + oi);
 #pragma unroll
           for (int j = 0; j < VW; ++j)
