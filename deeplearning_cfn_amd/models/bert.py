@@ -62,7 +62,15 @@ class SelfAttention(nn.Module):
         B, S, H = x.shape
         qkv = self.qkv(x).view(B, S, 3, self.heads, self.dh)
         q, k, v = (qkv[:, :, i].transpose(1, 2) for i in range(3))
-        o = F.scaled_dot_product_attention(q, k, v)
+        if q.is_cuda:
+            # Explicit batched-GEMM attention (rocBLAS bmm + fp32 softmax):
+            # no Triton / flash dispatch; S<=512 so the S^2 scores are tiny
+            # next to the MFMA projection GEMMs.
+            scores = (q @ k.transpose(-1, -2)) * (1.0 / math.sqrt(self.dh))
+            p = torch.softmax(scores.float(), dim=-1).to(q.dtype)
+            o = p @ v
+        else:
+            o = F.scaled_dot_product_attention(q, k, v)
         return self.proj(o.transpose(1, 2).reshape(B, S, H))
 
 

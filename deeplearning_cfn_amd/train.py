@@ -69,6 +69,11 @@ def build_argparser() -> argparse.ArgumentParser:
                    help="epochs between checkpoints (0 = off)")
     p.add_argument("--fault-step", type=int, default=None,
                    help="test-only: exit(13) at this step on DLX_FAULT_RANK")
+    p.add_argument("--profile", action="store_true",
+                   help="capture a per-op trace of --profile-steps steps to "
+                        "$EFS_MOUNT/trace_rank<r>.json (chrome://tracing); "
+                        "for kernel-level counters run under rocprofv3")
+    p.add_argument("--profile-steps", type=int, default=10)
     return p
 
 
@@ -147,6 +152,23 @@ def train(args: argparse.Namespace) -> dict:
     global_step = 0
     losses: list[float] = []
 
+    profiler = None
+    if args.profile:
+        # tracing is first-class.  torch.profiler rides
+        # roctracer on ROCm; the chrome trace lands on the shared mount so
+        # every rank's capture is inspectable from the master.
+        acts = [torch.profiler.ProfilerActivity.CPU]
+        if ctx.device.type == "cuda":
+            acts.append(torch.profiler.ProfilerActivity.CUDA)
+        trace_path = str(Path(efs) / f"trace_rank{ctx.rank}.json")
+        profiler = torch.profiler.profile(
+            activities=acts,
+            schedule=torch.profiler.schedule(
+                wait=1, warmup=2, active=args.profile_steps, repeat=1),
+            on_trace_ready=lambda p: p.export_chrome_trace(trace_path))
+        profiler.__enter__()
+        logger.log("profile", path=trace_path, steps=args.profile_steps)
+
     for epoch in range(start_epoch, args.num_epochs):
         model.train()
         timer.start()
@@ -167,6 +189,8 @@ def train(args: argparse.Namespace) -> dict:
             loss.backward()
             engine.finalize_backward()
             opt.step()
+            if profiler is not None:
+                profiler.step()
             global_step += 1
             interval_samples += args.batch_size
             if (args.fault_step is not None and ctx.rank == fault_rank
@@ -190,6 +214,8 @@ def train(args: argparse.Namespace) -> dict:
             logger.log("checkpoint", epoch=epoch + 1, prefix=prefix)
         ctx.barrier()
 
+    if profiler is not None:
+        profiler.__exit__(None, None, None)
     final_loss = losses[-1] if losses else float("nan")
     logger.log("done", final_loss=final_loss, steps=global_step)
     logger.close()
