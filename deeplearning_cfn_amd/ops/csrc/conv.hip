@@ -87,13 +87,12 @@ at::Tensor conv2d_dgrad_impl(const at::Tensor& g_in, const at::Tensor& w_in,
       for (int b = 0; b < 2; ++b)
         if (p.Kh <= ((a + p.pad) & 1) || p.Kw <= ((b + p.pad) & 1))
           any_empty = true;
-    auto dx = any_empty
-                  ? at::zeros({p.N, p.Ci, p.H, p.W},
-                              g.options().memory_format(
-                                  at::MemoryFormat::ChannelsLast))
-                  : at::empty({p.N, p.Ci, p.H, p.W},
-                              g.options().memory_format(
-                                  at::MemoryFormat::ChannelsLast));
+    // NB: at::zeros ignores TensorOptions.memory_format — allocate with
+    // empty (which honors it) and zero_() only when classes stay empty.
+    auto dx = at::empty({p.N, p.Ci, p.H, p.W},
+                        g.options().memory_format(
+                            at::MemoryFormat::ChannelsLast));
+    if (any_empty) dx.zero_();
     auto wt4 = wt.view({(long)p.Ci, (long)p.Kh, (long)p.Kw, (long)p.Co});
     for (int a = 0; a < 2; ++a) {
       for (int b = 0; b < 2; ++b) {
