@@ -51,7 +51,10 @@ constexpr int kBKw = 64;    // m per K-tile
 constexpr int kThreadsW = 256;  // 4 waves, 2x2, 64x64 per wave
 constexpr int kSlotsW = kBKw * 128 / 8;  // 16-B slots per operand tile
 
-// decompose slot f -> (mloc, c) of the image above (all shifts)
+// decompose slot f -> (mloc, c) of the image above (all shifts).
+// NCB = col-blocks per tile (8 for 128-wide, 16 for 256-wide operands);
+// LCB = log2(NCB).
+template <int LCB>
 DLX_DEV void slot_to_mc(int f, int& mloc, int& c) {
   const int e = f * 8;
   const int blk = e >> 6;       // 64-elem block
@@ -62,8 +65,8 @@ DLX_DEV void slot_to_mc(int f, int& mloc, int& c) {
   const int i = w >> 2;
   const int g = w & 3;
   const int kkcb = blk >> 3;
-  const int cb = kkcb & 7;
-  const int kk = kkcb >> 3;
+  const int cb = kkcb & ((1 << LCB) - 1);
+  const int kk = kkcb >> LCB;
   mloc = (kk << 5) + (g << 3) + (i << 2) + r;
   c = (cb << 4) + cc0;
 }
@@ -76,8 +79,10 @@ DLX_DEV void slot_to_mc(int f, int& mloc, int& c) {
 // column (l&15) across the 4 m-rows — the MFMA fragment gather.
 // Groups (l>>4) take consecutive blocks: lanes 0-31 cover 256 contiguous
 // bytes = all 64 LDS banks exactly once (conflict-free).
+template <int LCB>
 DLX_DEV int frag_off(int kk, int cb, int i, int lane) {
-  return (((kk * 8 + cb) * 8) + i * 4 + (lane >> 4)) * 64 + (lane & 15) * 4;
+  return (((kk << LCB) + cb) * 8 + i * 4 + (lane >> 4)) * 64 +
+         (lane & 15) * 4;
 }
 
 template <class AG, class BG>
@@ -118,7 +123,7 @@ __global__ __launch_bounds__(kThreadsW) void conv_wrw_kernel(
     for (int u = 0; u < kSlotsW / kThreadsW; ++u) {
       const int f = u * kThreadsW + threadIdx.x;
       int mloc, c;
-      slot_to_mc(f, mloc, c);
+      slot_to_mc<3>(f, mloc, c);
       const long m = mt0 + mloc;
       const uint16_t* sa =
           (m < mend) ? ag.src(m, row0 + c, zero_page) : zero_page;
@@ -147,15 +152,15 @@ __global__ __launch_bounds__(kThreadsW) void conv_wrw_kernel(
 #pragma unroll
       for (int mi = 0; mi < 4; ++mi) {
         const int cb = wr * 4 + mi;
-        bf16x4 lo = tr_read(la, frag_off(kk, cb, 0, lane));
-        bf16x4 hi = tr_read(la, frag_off(kk, cb, 1, lane));
+        bf16x4 lo = tr_read(la, frag_off<3>(kk, cb, 0, lane));
+        bf16x4 hi = tr_read(la, frag_off<3>(kk, cb, 1, lane));
         af[mi] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
       }
 #pragma unroll
       for (int ni = 0; ni < 4; ++ni) {
         const int cb = wc * 4 + ni;
-        bf16x4 lo = tr_read(lb, frag_off(kk, cb, 0, lane));
-        bf16x4 hi = tr_read(lb, frag_off(kk, cb, 1, lane));
+        bf16x4 lo = tr_read(lb, frag_off<3>(kk, cb, 0, lane));
+        bf16x4 hi = tr_read(lb, frag_off<3>(kk, cb, 1, lane));
         bfr[ni] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
       }
       __builtin_amdgcn_s_setprio(1);
@@ -192,6 +197,149 @@ __global__ __launch_bounds__(kThreadsW) void conv_wrw_kernel(
   }
 }
 
+// ---------------------------------------------------------------------
+// 256x256 8-phase wrw (Co >= 192): the igemm_kernel_8ph pipeline
+// (counted vmcnt, raw barriers, one LDS object, stage units = (operand,
+// k-half)) with tr-read fragments and the atomic fp32 epilogue.  Halves
+// the staged bytes per MAC vs the 128^2 tile — wrw is staging-BW-bound.
+// ---------------------------------------------------------------------
+
+template <class AG, class BG>
+__global__ __launch_bounds__(512) void conv_wrw_kernel_8ph(
+    AG ag, BG bg, float* __restrict__ out, const uint16_t* __restrict__ zp,
+    long Mred, long Rows, long Cols, int grid_n, int split,
+    long m_per_slice) {
+  constexpr int BMw = 256, BNw = 256, BKw = 64;
+  constexpr int THREADS = 512;
+  constexpr int SLOTS = BKw * BMw / 8;  // 2048 slots per operand
+  constexpr int MR = 8, NR = 4;         // per-wave 128(co) x 64(kf)
+
+  __shared__ uint16_t lds[2 * 2 * SLOTS * 8];  // [buf][A|B][slots][8]
+
+  int nwg = gridDim.x;
+  int bid = blockIdx.x;
+  {
+    int q = nwg / 8, r = nwg % 8;
+    int xcd = bid % 8, idx = bid / 8;
+    bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
+  }
+  const int tile = bid / split, slice = bid % split;
+  const int bm = tile / grid_n, bn = tile % grid_n;
+  const long row0 = (long)bm * BMw, col0 = (long)bn * BNw;
+  const long mbeg = slice * m_per_slice;
+  const long mend = (mbeg + m_per_slice < Mred) ? mbeg + m_per_slice : Mred;
+  if (mbeg >= mend) return;
+
+  const int lane = threadIdx.x % kWave;
+  const int wid = threadIdx.x / kWave;
+  const int wr = wid >> 2, wc = wid & 3;  // 2 x 4 waves
+  const int l15 = lane & 15, l4 = lane >> 4;
+
+  f32x4 acc[MR][NR] = {};
+  const long ntile = (mend - mbeg + BKw - 1) / BKw;
+
+  // stage unit u of tile t: u=0 A-kstep0, u=1 B-kstep0, u=2 A-ks1, u=3 B-ks1
+  auto stage_unit = [&](int buf, long t, int u) {
+    const long mt0 = mbeg + t * BKw;
+    const bool isB = u & 1;
+    const int half = u >> 1;
+    uint16_t* l0 = lds + (buf * 2 + (isB ? 1 : 0)) * SLOTS * 8;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int f = (half * 2 + i) * THREADS + threadIdx.x;
+      int mloc, c;
+      slot_to_mc<4>(f, mloc, c);
+      const long m = mt0 + mloc;
+      const uint16_t* src =
+          (m < mend) ? (isB ? bg.src(m, col0 + c, zp)
+                            : ag.src(m, row0 + c, zp))
+                     : zp;
+      __builtin_amdgcn_global_load_lds(
+          (glb_u32_t*)src, (lds_u32_t*)(l0 + (long)f * 8), 16, 0, 0);
+    }
+  };
+
+  // compute phase q (q>>1 = kstep, q&1 = mi half) of the tile in `buf`
+  auto phase_mfma = [&](int buf, int q) {
+    const uint16_t* la = lds + (buf * 2 + 0) * SLOTS * 8;
+    const uint16_t* lb = lds + (buf * 2 + 1) * SLOTS * 8;
+    const int ks = q >> 1;
+    const int mih = (q & 1) * 4;
+    bf16x8 af[4], bfr[NR];
+#pragma unroll
+    for (int i = 0; i < 4; ++i) {
+      const int cb = wr * 8 + mih + i;
+      bf16x4 lo = tr_read(la, frag_off<4>(ks, cb, 0, lane));
+      bf16x4 hi = tr_read(la, frag_off<4>(ks, cb, 1, lane));
+      af[i] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+    }
+#pragma unroll
+    for (int ni = 0; ni < NR; ++ni) {
+      const int cb = wc * 4 + ni;
+      bf16x4 lo = tr_read(lb, frag_off<4>(ks, cb, 0, lane));
+      bf16x4 hi = tr_read(lb, frag_off<4>(ks, cb, 1, lane));
+      bfr[ni] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
+    }
+    __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+    for (int i = 0; i < 4; ++i)
+#pragma unroll
+      for (int ni = 0; ni < NR; ++ni)
+        acc[mih + i][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            af[i], bfr[ni], acc[mih + i][ni], 0, 0, 0);
+    __builtin_amdgcn_s_setprio(0);
+  };
+
+#pragma unroll
+  for (int u = 0; u < 4; ++u) stage_unit(0, 0, u);
+
+  int buf = 0;
+  long t = 0;
+  for (; t + 1 < ntile; ++t) {
+    stage_unit(buf ^ 1, t + 1, 0);
+    asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    phase_mfma(buf, 0);
+    stage_unit(buf ^ 1, t + 1, 1);
+    __builtin_amdgcn_s_barrier();
+    phase_mfma(buf, 1);
+    stage_unit(buf ^ 1, t + 1, 2);
+    asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+    __builtin_amdgcn_s_barrier();
+    phase_mfma(buf, 2);
+    stage_unit(buf ^ 1, t + 1, 3);
+    __builtin_amdgcn_s_barrier();
+    phase_mfma(buf, 3);
+    buf ^= 1;
+  }
+  asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  phase_mfma(buf, 0);
+  phase_mfma(buf, 1);
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+  phase_mfma(buf, 2);
+  phase_mfma(buf, 3);
+
+#pragma unroll
+  for (int mi = 0; mi < MR; ++mi) {
+#pragma unroll
+    for (int ni = 0; ni < NR; ++ni) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long row = row0 + wr * 128 + mi * 16 + l4 * 4 + r;
+        const long col = col0 + wc * 64 + ni * 16 + l15;
+        if (row < Rows && col < Cols) {
+          if (split > 1)
+            unsafeAtomicAdd(&out[row * Cols + col], acc[mi][ni][r]);
+          else
+            out[row * Cols + col] = acc[mi][ni][r];
+        }
+      }
+    }
+  }
+}
+
 }  // namespace
 
 // g: [N,Co,Ho,Wo] CL bf16;  x: [N,Ci,H,W] CL bf16 ->
@@ -214,10 +362,14 @@ at::Tensor conv2d_wrw_impl(const at::Tensor& g_in, const at::Tensor& x_in,
   TORCH_CHECK(p.Co % 8 == 0, "wrw: Cout % 8 != 0 (use library path)");
 
   const long Rows = p.Co, Cols = p.K;
-  const int gm = (int)((Rows + kBMw - 1) / kBMw);
-  const int gn = (int)((Cols + kBNw - 1) / kBNw);
-  // split the m reduction so the grid reaches ~2 blocks/CU
-  long split = 768 / ((long)gm * gn);
+  // big-Co shapes: 256^2 8-phase tile (half the staged bytes per MAC,
+  // counted-vmcnt pipeline); small Co: 128^2 4-wave tile
+  const bool big = Rows >= 192 && Cols >= 192;
+  const int BM = big ? 256 : kBMw, BN = big ? 256 : kBNw;
+  const int gm = (int)((Rows + BM - 1) / BM);
+  const int gn = (int)((Cols + BN - 1) / BN);
+  // split the m reduction to fill the chip (~1-2 blocks/CU)
+  long split = (big ? 448 : 768) / ((long)gm * gn);
   const long max_split = (p.M + kBKw - 1) / kBKw;
   if (split > max_split) split = max_split;
   if (split < 1) split = 1;
@@ -238,11 +390,18 @@ at::Tensor conv2d_wrw_impl(const at::Tensor& g_in, const at::Tensor& x_in,
                (long)p.Co};
   ConvFwdGather bg{reinterpret_cast<const uint16_t*>(x.data_ptr()), p};
   const int nblocks = gm * gn * (int)split;
-  conv_wrw_kernel<RowGather, ConvFwdGather>
-      <<<nblocks, kThreadsW, 0, s>>>(
-          ag, bg, outf.data_ptr<float>(),
-          reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, Rows, Cols,
-          gn, (int)split, m_per_slice);
+  if (big)
+    conv_wrw_kernel_8ph<RowGather, ConvFwdGather>
+        <<<nblocks, 512, 0, s>>>(
+            ag, bg, outf.data_ptr<float>(),
+            reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, Rows,
+            Cols, gn, (int)split, m_per_slice);
+  else
+    conv_wrw_kernel<RowGather, ConvFwdGather>
+        <<<nblocks, kThreadsW, 0, s>>>(
+            ag, bg, outf.data_ptr<float>(),
+            reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, Rows,
+            Cols, gn, (int)split, m_per_slice);
 
   // [Co, Kf] fp32 -> bf16 tensor with channels_last strides of the weight
   return outf.to(at::kBFloat16)
