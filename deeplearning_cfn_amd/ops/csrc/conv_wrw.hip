@@ -362,9 +362,14 @@ at::Tensor conv2d_wrw_impl(const at::Tensor& g_in, const at::Tensor& x_in,
   TORCH_CHECK(p.Co % 8 == 0, "wrw: Cout % 8 != 0 (use library path)");
 
   const long Rows = p.Co, Cols = p.K;
-  // big-Co shapes: 256^2 8-phase tile (half the staged bytes per MAC,
-  // counted-vmcnt pipeline); small Co: 128^2 4-wave tile
-  const bool big = Rows >= 192 && Cols >= 192;
+  // Measured (profiles/): the 128^2 4-wave tile beats the 256^2 8-phase
+  // variant on every ResNet wrw shape (occupancy/tail balance outweighs
+  // the halved staged bytes) — keep 8ph behind an env flag for re-tests.
+  static const bool want8ph = []() {
+    const char* e = getenv("DLX_WRW_8PH");
+    return e && e[0] == '1';
+  }();
+  const bool big = want8ph && Rows >= 192 && Cols >= 192;
   const int BM = big ? 256 : kBMw, BN = big ? 256 : kBNw;
   const int gm = (int)((Rows + BM - 1) / BM);
   const int gn = (int)((Cols + BN - 1) / BN);
