@@ -17,10 +17,14 @@ def summarize(db_path: str, out_path: str, title: str = "") -> None:
     sym = next(t for t in names if t.startswith("rocpd_info_kernel"))
     disp = next(t for t in names if t.startswith("rocpd_kernel_dispatch"))
     rows = list(c.execute(f"""
-        SELECT s.display_name, COUNT(*), SUM(d.end-d.start)/1e6,
+        SELECT s.kernel_name, COUNT(*), SUM(d.end-d.start)/1e6,
                AVG(d.end-d.start)/1e3
         FROM {disp} d JOIN {sym} s ON d.kernel_id = s.id
-        GROUP BY s.display_name ORDER BY SUM(d.end-d.start) DESC"""))
+        GROUP BY s.kernel_name ORDER BY SUM(d.end-d.start) DESC"""))
+    import subprocess
+    dm = subprocess.run(["c++filt"], input="\n".join(r[0] for r in rows),
+                        capture_output=True, text=True).stdout.splitlines()
+    rows = [(d,) + tuple(r[1:]) for r, d in zip(rows, dm)]
     total = sum(r[2] for r in rows)
     out = Path(out_path)
     out.parent.mkdir(parents=True, exist_ok=True)
