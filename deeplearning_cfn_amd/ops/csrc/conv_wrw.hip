@@ -414,4 +414,42 @@ at::Tensor conv2d_wrw_impl(const at::Tensor& g_in, const at::Tensor& x_in,
       .permute({0, 3, 1, 2});
 }
 
+// out[n,k] = sum_m A[m,n] * B[m,k]  (A^T·B, reduction over the leading
+// axis of both operands — the Linear weight-grad dW = dY^T·X).  Reuses
+// the wrw reduce-major kernel (tr-read fragments, split-m atomics)
+// instead of materialising two transposed copies for gemm_nt.
+at::Tensor gemm_tn_mred_impl(const at::Tensor& a_in, const at::Tensor& b_in) {
+  auto A = a_in.contiguous();
+  auto B = b_in.contiguous();
+  TORCH_CHECK(A.scalar_type() == at::kBFloat16, "gemm_tn: bf16 only");
+  TORCH_CHECK(A.size(0) == B.size(0), "gemm_tn: reduction extents differ");
+  const long M = A.size(0), Rows = A.size(1), Cols = B.size(1);
+  TORCH_CHECK(Rows % 8 == 0 && Cols % 8 == 0, "gemm_tn: dims % 8 != 0");
+
+  const int gm = (int)((Rows + kBMw - 1) / kBMw);
+  const int gn = (int)((Cols + kBNw - 1) / kBNw);
+  long split = 768 / ((long)gm * gn);
+  const long max_split = (M + kBKw - 1) / kBKw;
+  if (split > max_split) split = max_split;
+  if (split < 1) split = 1;
+  long m_per_slice = (M + split - 1) / split;
+  m_per_slice = ((m_per_slice + kBKw - 1) / kBKw) * kBKw;
+
+  auto outf = (split > 1)
+                  ? at::zeros({Rows, Cols}, A.options().dtype(at::kFloat))
+                  : at::empty({Rows, Cols}, A.options().dtype(at::kFloat));
+  static at::Tensor zp;
+  if (!zp.defined() || zp.device() != A.device())
+    zp = at::zeros({16}, A.options().dtype(at::kBFloat16));
+  auto s = at::cuda::getCurrentCUDAStream();
+  RowGather ag{reinterpret_cast<const uint16_t*>(A.data_ptr()), M, Rows};
+  RowGather bg{reinterpret_cast<const uint16_t*>(B.data_ptr()), M, Cols};
+  conv_wrw_kernel<RowGather, RowGather>
+      <<<gm * gn * (int)split, kThreadsW, 0, s>>>(
+          ag, bg, outf.data_ptr<float>(),
+          reinterpret_cast<const uint16_t*>(zp.data_ptr()), M, Rows, Cols,
+          gn, (int)split, m_per_slice);
+  return outf.to(at::kBFloat16);
+}
+
 }  // namespace dlx

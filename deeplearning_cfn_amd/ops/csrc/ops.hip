@@ -216,9 +216,11 @@ at::Tensor gemm_nn(const at::Tensor& g, const at::Tensor& w) {
   return gemm_nt_impl(g, w.t().contiguous(), true);
 }
 
-// g[M,N]^T @ x[M,K] -> [N,K]  (wgrad)
+// g[M,N]^T @ x[M,K] -> [N,K]  (wgrad) — reduce-major kernel, no
+// transposed copies (conv_wrw.hip)
+at::Tensor gemm_tn_mred_impl(const at::Tensor& a, const at::Tensor& b);
 at::Tensor gemm_tn(const at::Tensor& g, const at::Tensor& x) {
-  return gemm_nt_impl(g.t().contiguous(), x.t().contiguous(), true);
+  return gemm_tn_mred_impl(g, x);
 }
 
 }  // namespace
