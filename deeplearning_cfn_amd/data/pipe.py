@@ -47,8 +47,16 @@ class HostPinnedPrefetcher:
     def _produce(self) -> None:
         try:
             while not self._stop.is_set():
-                batch = next(self.source)
-                self._q.put(self._pin(batch))
+                batch = self._pin(next(self.source))
+                # bounded put so close() can always stop the thread (a
+                # daemon thread killed mid-HIP-call aborts the process
+                # at interpreter teardown)
+                while not self._stop.is_set():
+                    try:
+                        self._q.put(batch, timeout=0.1)
+                        break
+                    except queue.Full:
+                        continue
         except StopIteration:
             self._q.put(None)
 
@@ -82,3 +90,5 @@ class HostPinnedPrefetcher:
                 self._q.get_nowait()
             except queue.Empty:
                 break
+        if self.cuda and self._thread.is_alive():
+            self._thread.join(timeout=2.0)

@@ -141,8 +141,10 @@ def train(args: argparse.Namespace) -> dict:
                                  rank=ctx.rank, dtype=dtype,
                                  channels_last=True,
                                  pinned=ctx.device.type == "cuda")
+    prefetcher = None
     if ctx.device.type == "cuda" and not args.no_prefetch:
-        batches = iter(HostPinnedPrefetcher(source, ctx.device))
+        prefetcher = HostPinnedPrefetcher(source, ctx.device)
+        batches = iter(prefetcher)
     else:
         batches = iter(source)
 
@@ -216,6 +218,8 @@ def train(args: argparse.Namespace) -> dict:
 
     if profiler is not None:
         profiler.__exit__(None, None, None)
+    if prefetcher is not None:
+        prefetcher.close()
     final_loss = losses[-1] if losses else float("nan")
     logger.log("done", final_loss=final_loss, steps=global_step)
     logger.close()
