@@ -455,6 +455,13 @@ inline void igemm_launch_om(const AG& ag, const BG& bg, void* cptr,
     const int gm = (int)((M + 255) / 256), gn = (int)((Nc + 255) / 256);
     igemm_kernel_8ph<AG, BG, OUT_BF16, OM>
         <<<gm * gn, 512, 0, s>>>(ag, bg, cptr, zp, M, Nc, K, gn, om);
+  } else if (Nc <= 96 && M >= 1024) {
+    // narrow-N (layer1-class: N = Ci/Co = 64): a 128x128 tile wastes
+    // half its MFMA work on the zero page — use a 256x64 tile (4 waves
+    // of 64x64, 80 KiB LDS, 2 blocks/CU)
+    const int gm = (int)((M + 255) / 256), gn = (int)((Nc + 63) / 64);
+    igemm_kernel<256, 64, 64, 4, 1, AG, BG, OUT_BF16, OM>
+        <<<gm * gn, 256, 0, s>>>(ag, bg, cptr, zp, M, Nc, K, gn, om);
   } else {
     const int gm = (int)((M + 127) / 128), gn = (int)((Nc + 127) / 128);
     igemm_kernel<128, 128, 64, 2, 2, AG, BG, OUT_BF16, OM>
