@@ -95,6 +95,9 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> ln_bwd_impl(
 at::Tensor gelu_fwd_impl(const at::Tensor& x);
 at::Tensor gelu_bwd_impl(const at::Tensor& g, const at::Tensor& x);
 
+// conv_wrw.hip — reduce-major GEMM (Linear wgrad)
+at::Tensor gemm_tn_mred_impl(const at::Tensor& a, const at::Tensor& b);
+
 namespace {
 
 hipStream_t cur_stream() { return at::cuda::getCurrentCUDAStream(); }
@@ -218,7 +221,6 @@ at::Tensor gemm_nn(const at::Tensor& g, const at::Tensor& w) {
 
 // g[M,N]^T @ x[M,K] -> [N,K]  (wgrad) — reduce-major kernel, no
 // transposed copies (conv_wrw.hip)
-at::Tensor gemm_tn_mred_impl(const at::Tensor& a, const at::Tensor& b);
 at::Tensor gemm_tn(const at::Tensor& g, const at::Tensor& x) {
   return gemm_tn_mred_impl(g, x);
 }
