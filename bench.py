@@ -34,8 +34,9 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=10)
-    p.add_argument("--batch-size", type=int, default=256,
-                   help="per-GPU batch (weak scaling)")
+    p.add_argument("--batch-size", type=int, default=1024,
+                   help="per-GPU batch (weak scaling; 1024 sizes the "
+                        "step for 288 GB HBM3E — measured fastest)")
     p.add_argument("--num-layers", type=int, default=50)
     p.add_argument("--dtype", default="bf16")
     p.add_argument("--impl", default="auto")
