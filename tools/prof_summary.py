@@ -34,7 +34,8 @@ def summarize(db_path: str, out_path: str, title: str = "") -> None:
                 f"{total:.1f} ms over {sum(r[1] for r in rows)} dispatches\n\n")
         f.write("| kernel | calls | total ms | avg µs | % |\n|---|---|---|---|---|\n")
         for name, n, ms, avg_us in rows[:40]:
-            short = name.split("(")[0][:100]
+            short = name.replace("(anonymous namespace)::", "")
+            short = short.split("(")[0][:110]
             f.write(f"| `{short}` | {n} | {ms:.2f} | {avg_us:.1f} | "
                     f"{100*ms/total:.1f} |\n")
     print(f"wrote {out} ({len(rows)} kernels, {total:.1f} ms)")
