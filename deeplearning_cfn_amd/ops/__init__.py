@@ -53,6 +53,26 @@ def native_available() -> bool:
     return _EXT is not None
 
 
+class _DebugSyncOps:
+    """DLX_DEBUG_SYNC=1: synchronize after every native op so an async
+    kernel fault (page fault, XID) surfaces at the faulting call site
+    instead of a later unrelated one (SURVEY.md §6.2 race/sanitizer
+    plan: the cheap device-side invariant)."""
+
+    def __init__(self, ops):
+        self._ops = ops
+
+    def __getattr__(self, name):
+        fn = getattr(self._ops, name)
+
+        def wrapped(*a, **k):
+            r = fn(*a, **k)
+            torch.cuda.synchronize()
+            return r
+
+        return wrapped
+
+
 def native():
     """The loaded torch.ops namespace; raises loudly if unavailable on GPU."""
     _try_load()
@@ -61,6 +81,8 @@ def native():
             f"dlx native HIP extension unavailable: {_EXT_ERR}. "
             f"On a GPU box this is fatal (no silent torch fallback); set "
             f"DLX_IMPL=torch to run the reference-parity fallback explicitly.")
+    if os.environ.get("DLX_DEBUG_SYNC") == "1":
+        return _DebugSyncOps(_EXT)
     return _EXT
 
 
