@@ -25,6 +25,16 @@ cfn-template/deeplearning.template resources, examples/mxnet,
 examples/tensorflow) with confidence tags.
 """
 
+import os as _os
+
+# Only the 3-channel stem conv still routes through MIOpen; its default
+# hybrid find mode spends tens of GPU-seconds benchmarking candidate
+# kernels per process (measured 52 s at batch 1024).  FAST picks an
+# immediate-mode kernel — the stem is ~1%% of the step, the find cost is
+# not.  Users can override by exporting MIOPEN_FIND_MODE themselves.
+_os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+
+
 __version__ = "0.1.0"
 
 from . import contract  # noqa: F401
