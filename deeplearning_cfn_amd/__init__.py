@@ -25,14 +25,11 @@ cfn-template/deeplearning.template resources, examples/mxnet,
 examples/tensorflow) with confidence tags.
 """
 
-import os as _os
-
-# Only the 3-channel stem conv still routes through MIOpen; its default
-# hybrid find mode spends tens of GPU-seconds benchmarking candidate
-# kernels per process (measured 52 s at batch 1024).  FAST picks an
-# immediate-mode kernel — the stem is ~1%% of the step, the find cost is
-# not.  Users can override by exporting MIOPEN_FIND_MODE themselves.
-_os.environ.setdefault("MIOPEN_FIND_MODE", "FAST")
+# NOTE: only the 3-channel stem conv still routes through MIOpen.  Its
+# default (hybrid) find mode spends ~50 GPU-seconds benchmarking stem
+# candidates once per process at batch 1024 — but MIOPEN_FIND_MODE=FAST
+# was measured to select a naive kernel ~500x slower per step, so the
+# one-time find cost is the right trade and no override is set here.
 
 
 __version__ = "0.1.0"
