@@ -49,8 +49,17 @@ at::Tensor conv2d_fwd_impl(const at::Tensor& x_in, const at::Tensor& w_in,
                      x.options().memory_format(at::MemoryFormat::ChannelsLast));
   auto zp = zero_page_for(x);
   auto s = at::cuda::getCurrentCUDAStream();
-  ConvFwdGather ag{reinterpret_cast<const uint16_t*>(x.data_ptr()), p};
   RowGather bg{reinterpret_cast<const uint16_t*>(w.data_ptr()), p.Co, p.K};
+  if (p.Kh == 1 && p.Kw == 1 && p.stride == 1 && p.pad == 0) {
+    // 1x1/s1 is a plain GEMM: y[m,co] = x[m,:] @ W[co,:]^T — skip the
+    // im2col gather's per-slot address math entirely
+    RowGather ag{reinterpret_cast<const uint16_t*>(x.data_ptr()), p.M, p.K};
+    igemm_launch<RowGather, RowGather, true>(
+        ag, bg, y.data_ptr(),
+        reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, p.Co, p.K, s);
+    return y;
+  }
+  ConvFwdGather ag{reinterpret_cast<const uint16_t*>(x.data_ptr()), p};
   igemm_launch<ConvFwdGather, RowGather, true>(
       ag, bg, y.data_ptr(),
       reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, p.Co, p.K, s);
@@ -133,8 +142,16 @@ at::Tensor conv2d_dgrad_impl(const at::Tensor& g_in, const at::Tensor& w_in,
 
   auto dx = at::empty({p.N, p.Ci, p.H, p.W},
                       g.options().memory_format(at::MemoryFormat::ChannelsLast));
-  ConvDgradGather ag{reinterpret_cast<const uint16_t*>(g.data_ptr()), p};
   RowGather bg{reinterpret_cast<const uint16_t*>(wt.data_ptr()), p.Ci, p.K};
+  if (p.Kh == 1 && p.Kw == 1 && p.stride == 1 && p.pad == 0) {
+    // 1x1/s1 dgrad is a plain GEMM: dx[m,ci] = g[m,:] @ Wt[ci,:]^T
+    RowGather ag{reinterpret_cast<const uint16_t*>(g.data_ptr()), p.M, p.K};
+    igemm_launch<RowGather, RowGather, true>(
+        ag, bg, dx.data_ptr(),
+        reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, p.Ci, p.K, s);
+    return dx;
+  }
+  ConvDgradGather ag{reinterpret_cast<const uint16_t*>(g.data_ptr()), p};
   igemm_launch<ConvDgradGather, RowGather, true>(
       ag, bg, dx.data_ptr(),
       reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, p.Ci, p.K, s);

@@ -85,11 +85,14 @@ DLX_DEV int frag_off(int kk, int cb, int i, int lane) {
          (lane & 15) * 4;
 }
 
-template <class AG, class BG>
-__global__ __launch_bounds__(kThreadsW) void conv_wrw_kernel(
+template <class AG, class BG, int WM = 2, int WN = 2>
+__global__ __launch_bounds__(WM* WN * 64) void conv_wrw_kernel(
     AG ag, BG bg, float* __restrict__ out,  // [BM-rows=co, BN-cols=kf]
     const uint16_t* __restrict__ zero_page, long Mred, long Rows, long Cols,
     int grid_n, int split, long m_per_slice) {
+  constexpr int THREADS = WM * WN * 64;
+  constexpr int MR = (128 / WM) / 16;
+  constexpr int NR = (128 / WN) / 16;
   __shared__ uint16_t lds[2 * 2 * kSlotsW * 8];  // [buf][A|B][slots][8]
 
   int nwg = gridDim.x;
@@ -109,10 +112,10 @@ __global__ __launch_bounds__(kThreadsW) void conv_wrw_kernel(
 
   const int lane = threadIdx.x % kWave;
   const int wid = threadIdx.x / kWave;
-  const int wr = wid >> 1, wc = wid & 1;  // 2x2 waves, 64x64 each
+  const int wr = wid / WN, wc = wid % WN;
   const int l15 = lane & 15, l4 = lane >> 4;
 
-  f32x4 acc[4][4] = {};
+  f32x4 acc[MR][NR] = {};
   const long ntile = (mend - mbeg + kBKw - 1) / kBKw;
 
   auto stage = [&](int buf, long t) {
@@ -120,8 +123,8 @@ __global__ __launch_bounds__(kThreadsW) void conv_wrw_kernel(
     uint16_t* la = lds + (buf * 2 + 0) * kSlotsW * 8;
     uint16_t* lb = lds + (buf * 2 + 1) * kSlotsW * 8;
 #pragma unroll
-    for (int u = 0; u < kSlotsW / kThreadsW; ++u) {
-      const int f = u * kThreadsW + threadIdx.x;
+    for (int u = 0; u < kSlotsW / THREADS; ++u) {
+      const int f = u * THREADS + threadIdx.x;
       int mloc, c;
       slot_to_mc<3>(f, mloc, c);
       const long m = mt0 + mloc;
@@ -147,27 +150,27 @@ __global__ __launch_bounds__(kThreadsW) void conv_wrw_kernel(
     const uint16_t* lb = lds + (buf * 2 + 1) * kSlotsW * 8;
 #pragma unroll
     for (int kk = 0; kk < kBKw / 32; ++kk) {
-      // A fragments: rows (co) of this wave's 64-row strip
-      bf16x8 af[4], bfr[4];
+      // A fragments: rows (co) of this wave's strip
+      bf16x8 af[MR], bfr[NR];
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi) {
-        const int cb = wr * 4 + mi;
+      for (int mi = 0; mi < MR; ++mi) {
+        const int cb = wr * MR + mi;
         bf16x4 lo = tr_read(la, frag_off<3>(kk, cb, 0, lane));
         bf16x4 hi = tr_read(la, frag_off<3>(kk, cb, 1, lane));
         af[mi] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
       }
 #pragma unroll
-      for (int ni = 0; ni < 4; ++ni) {
-        const int cb = wc * 4 + ni;
+      for (int ni = 0; ni < NR; ++ni) {
+        const int cb = wc * NR + ni;
         bf16x4 lo = tr_read(lb, frag_off<3>(kk, cb, 0, lane));
         bf16x4 hi = tr_read(lb, frag_off<3>(kk, cb, 1, lane));
         bfr[ni] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
       }
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int mi = 0; mi < 4; ++mi)
+      for (int mi = 0; mi < MR; ++mi)
 #pragma unroll
-        for (int ni = 0; ni < 4; ++ni)
+        for (int ni = 0; ni < NR; ++ni)
           acc[mi][ni] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               af[mi], bfr[ni], acc[mi][ni], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
@@ -179,13 +182,13 @@ __global__ __launch_bounds__(kThreadsW) void conv_wrw_kernel(
 
   // epilogue: D map col=lane&15, row=(lane>>4)*4+r; atomic accumulate
 #pragma unroll
-  for (int mi = 0; mi < 4; ++mi) {
+  for (int mi = 0; mi < MR; ++mi) {
 #pragma unroll
-    for (int ni = 0; ni < 4; ++ni) {
+    for (int ni = 0; ni < NR; ++ni) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const long row = row0 + wr * 64 + mi * 16 + l4 * 4 + r;
-        const long col = col0 + wc * 64 + ni * 16 + l15;
+        const long row = row0 + wr * (128 / WM) + mi * 16 + l4 * 4 + r;
+        const long col = col0 + wc * (128 / WN) + ni * 16 + l15;
         if (row < Rows && col < Cols) {
           if (split > 1)
             unsafeAtomicAdd(&out[row * Cols + col], acc[mi][ni][r]);
@@ -395,8 +398,18 @@ at::Tensor conv2d_wrw_impl(const at::Tensor& g_in, const at::Tensor& x_in,
                (long)p.Co};
   ConvFwdGather bg{reinterpret_cast<const uint16_t*>(x.data_ptr()), p};
   const int nblocks = gm * gn * (int)split;
+  static const bool w8 = []() {
+    const char* e = getenv("DLX_WRW_WAVES");
+    return e && e[0] == '8';
+  }();
   if (big)
     conv_wrw_kernel_8ph<RowGather, ConvFwdGather>
+        <<<nblocks, 512, 0, s>>>(
+            ag, bg, outf.data_ptr<float>(),
+            reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, Rows,
+            Cols, gn, (int)split, m_per_slice);
+  else if (w8)
+    conv_wrw_kernel<RowGather, ConvFwdGather, 2, 4>
         <<<nblocks, 512, 0, s>>>(
             ag, bg, outf.data_ptr<float>(),
             reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, Rows,
