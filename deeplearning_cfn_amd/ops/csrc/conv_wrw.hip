@@ -398,9 +398,10 @@ at::Tensor conv2d_wrw_impl(const at::Tensor& g_in, const at::Tensor& x_in,
                (long)p.Co};
   ConvFwdGather bg{reinterpret_cast<const uint16_t*>(x.data_ptr()), p};
   const int nblocks = gm * gn * (int)split;
+  // 8 waves (2x4) measured +11%% over 4 waves at b1024 (profiles/)
   static const bool w8 = []() {
     const char* e = getenv("DLX_WRW_WAVES");
-    return e && e[0] == '8';
+    return !(e && e[0] == '4');
   }();
   if (big)
     conv_wrw_kernel_8ph<RowGather, ConvFwdGather>
