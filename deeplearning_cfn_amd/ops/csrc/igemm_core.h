@@ -18,6 +18,8 @@
 // vmcnt(0)+barrier per K-tile (guide §5.5 T3 minimum 2-phase).
 #pragma once
 
+#include <cstdlib>
+
 #include "dlx_common.h"
 
 namespace dlx {
@@ -463,9 +465,17 @@ inline void igemm_launch_om(const AG& ag, const BG& bg, void* cptr,
     igemm_kernel<256, 64, 64, 4, 1, AG, BG, OUT_BF16, OM>
         <<<gm * gn, 256, 0, s>>>(ag, bg, cptr, zp, M, Nc, K, gn, om);
   } else {
+    static const bool w8 = []() {
+      const char* e = getenv("DLX_IGEMM_W8");
+      return e && e[0] == '1';
+    }();
     const int gm = (int)((M + 127) / 128), gn = (int)((Nc + 127) / 128);
-    igemm_kernel<128, 128, 64, 2, 2, AG, BG, OUT_BF16, OM>
-        <<<gm * gn, 256, 0, s>>>(ag, bg, cptr, zp, M, Nc, K, gn, om);
+    if (w8)
+      igemm_kernel<128, 128, 64, 2, 4, AG, BG, OUT_BF16, OM>
+          <<<gm * gn, 512, 0, s>>>(ag, bg, cptr, zp, M, Nc, K, gn, om);
+    else
+      igemm_kernel<128, 128, 64, 2, 2, AG, BG, OUT_BF16, OM>
+          <<<gm * gn, 256, 0, s>>>(ag, bg, cptr, zp, M, Nc, K, gn, om);
   }
 }
 
