@@ -166,12 +166,14 @@ __global__ void bn_eval_coeffs_kernel(const float* __restrict__ w,
 // ---- normalize: y = x*scale[c] + shift[c] (+ReLU) ----------------------
 // Same chunk/stripe geometry; scale/shift hoisted out of the row loop.
 
-template <typename T, bool RELU, bool VEC, bool RES = false>
+template <typename T, bool RELU, bool VEC, bool RES = false,
+          bool WMASK = false>
 __global__ void bn_norm_kernel(const T* __restrict__ x, T* __restrict__ y,
                                const float* __restrict__ scale,
                                const float* __restrict__ shift, long R, int C,
                                int CT, long rows_per_stripe,
-                               const T* __restrict__ res = nullptr) {
+                               const T* __restrict__ res = nullptr,
+                               uint8_t* __restrict__ mask = nullptr) {
   using VT = VecTraits<T>;
   using Vec = typename VT::Vec;
   constexpr int VW = VEC ? VT::width : 1;
@@ -194,13 +196,18 @@ __global__ void bn_norm_kernel(const T* __restrict__ x, T* __restrict__ y,
       Vec rv;
       if (RES) rv = *reinterpret_cast<const Vec*>(res + i);
       Vec yv;
+      uint8_t mb = 0;
 #pragma unroll
       for (int j = 0; j < VW; ++j) {
         float v = VT::load(xv, j) * sc[j] + sh[j];
         if (RES) v += VT::load(rv, j);
+        if (WMASK && v > 0.f) mb |= (uint8_t)(1u << j);
         VT::store(yv, j, RELU ? fmaxf(v, 0.f) : v);
       }
       *reinterpret_cast<Vec*>(y + i) = yv;
+      // relu bit-mask: 1 byte per VW-chunk — lets backward skip re-reading
+      // y (saves ~2 full tensor passes per BN backward)
+      if (WMASK) mask[r * (C / VW) + blockIdx.y * CT + chunk] = mb;
     } else {
       float v = VT::load1(x + i) * sc[0] + sh[0];
       if (RES) v += VT::load1(res + i);
@@ -211,14 +218,16 @@ __global__ void bn_norm_kernel(const T* __restrict__ x, T* __restrict__ y,
 
 // ---- backward reduce: sum(dy), sum(dy * xhat) per channel --------------
 
-template <typename T, bool RELU, bool VEC>
+template <typename T, bool RELU, bool VEC, bool MASKED = false>
 __global__ void bn_reduce_bwd_kernel(const T* __restrict__ dy,
                                      const T* __restrict__ x,
                                      const T* __restrict__ y,
                                      const float* __restrict__ save_mean,
                                      const float* __restrict__ save_invstd,
                                      float* __restrict__ sums, long R, int C,
-                                     int CT, long rows_per_stripe) {
+                                     int CT, long rows_per_stripe,
+                                     const uint8_t* __restrict__ mask =
+                                         nullptr) {
   using VT = VecTraits<T>;
   using Vec = typename VT::Vec;
   constexpr int VW = VEC ? VT::width : 1;
@@ -245,11 +254,17 @@ __global__ void bn_reduce_bwd_kernel(const T* __restrict__ dy,
       Vec gv = *reinterpret_cast<const Vec*>(dy + i);
       Vec xv = *reinterpret_cast<const Vec*>(x + i);
       Vec yv;
-      if (RELU) yv = *reinterpret_cast<const Vec*>(y + i);
+      uint8_t mb = 0xff;
+      if (MASKED) mb = mask[r * (C / VW) + blockIdx.y * CT + chunk];
+      else if (RELU) yv = *reinterpret_cast<const Vec*>(y + i);
 #pragma unroll
       for (int j = 0; j < VW; ++j) {
         float g = VT::load(gv, j);
-        if (RELU && VT::load(yv, j) <= 0.f) g = 0.f;
+        if (MASKED) {
+          if (!(mb & (1u << j))) g = 0.f;
+        } else if (RELU && VT::load(yv, j) <= 0.f) {
+          g = 0.f;
+        }
         float xh = (VT::load(xv, j) - mean[j]) * invstd[j];
         a[j] += g;
         bx[j] += g * xh;
@@ -314,14 +329,16 @@ __global__ void bn_finalize_bwd_kernel(const float* __restrict__ sums,
 
 // ---- dx pass: dx = cA*dy_eff + cB*x + cD (coefficients hoisted) --------
 
-template <typename T, bool RELU, bool VEC, bool DRES = false>
+template <typename T, bool RELU, bool VEC, bool DRES = false,
+          bool MASKED = false>
 __global__ void bn_dx_kernel(const T* __restrict__ dy, const T* __restrict__ x,
                              const T* __restrict__ y,
                              const float* __restrict__ cA,
                              const float* __restrict__ cB,
                              const float* __restrict__ cD, T* __restrict__ dx,
                              long R, int C, int CT, long rows_per_stripe,
-                             T* __restrict__ dres = nullptr) {
+                             T* __restrict__ dres = nullptr,
+                             const uint8_t* __restrict__ mask = nullptr) {
   using VT = VecTraits<T>;
   using Vec = typename VT::Vec;
   constexpr int VW = VEC ? VT::width : 1;
@@ -344,12 +361,18 @@ __global__ void bn_dx_kernel(const T* __restrict__ dy, const T* __restrict__ x,
       Vec gv = *reinterpret_cast<const Vec*>(dy + i);
       Vec xv = *reinterpret_cast<const Vec*>(x + i);
       Vec yv;
-      if (RELU) yv = *reinterpret_cast<const Vec*>(y + i);
+      uint8_t mb = 0xff;
+      if (MASKED) mb = mask[r * (C / VW) + blockIdx.y * CT + chunk];
+      else if (RELU) yv = *reinterpret_cast<const Vec*>(y + i);
       Vec ov, rv;
 #pragma unroll
       for (int j = 0; j < VW; ++j) {
         float g = VT::load(gv, j);
-        if (RELU && VT::load(yv, j) <= 0.f) g = 0.f;
+        if (MASKED) {
+          if (!(mb & (1u << j))) g = 0.f;
+        } else if (RELU && VT::load(yv, j) <= 0.f) {
+          g = 0.f;
+        }
         if (DRES) VT::store(rv, j, g);
         VT::store(ov, j, A[j] * g + B[j] * VT::load(xv, j) + D[j]);
       }
@@ -377,7 +400,8 @@ static T* tpm(at::Tensor& t) {
 
 }  // namespace
 
-std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd_train_impl(
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
+bn_fwd_train_impl(
     const at::Tensor& x_in, const at::Tensor& w, const at::Tensor& b,
     at::Tensor& rmean, at::Tensor& rvar, double momentum, double eps,
     bool relu) {
@@ -395,6 +419,12 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd_train_impl(
   auto scale = at::empty({sh.C}, fopts);
   auto shift = at::empty({sh.C}, fopts);
   auto y = at::empty_like(x);
+  // relu bit-mask (1 byte per VW-chunk) — backward then never re-reads y
+  const bool wmask = relu && sh.vec;
+  const int vw = bf ? 8 : 4;
+  auto mask = wmask ? at::empty({sh.R * (sh.C / vw)},
+                                x.options().dtype(at::kByte))
+                    : at::empty({0}, x.options().dtype(at::kByte));
 
 #define DLX_BN_T(T, VEC)                                                     \
   {                                                                          \
@@ -407,7 +437,13 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd_train_impl(
         save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),          \
         scale.data_ptr<float>(), shift.data_ptr<float>(), sh.R, sh.C,        \
         sh.stripes, (float)momentum, (float)eps);                            \
-    if (relu)                                                                \
+    if (relu && VEC)                                                         \
+      bn_norm_kernel<T, true, VEC, false, true>                              \
+          <<<sh.grid, kBlock, 0, s>>>(                                       \
+              tp<T>(x), tpm<T>(y), scale.data_ptr<float>(),                  \
+              shift.data_ptr<float>(), sh.R, sh.C, sh.CT,                    \
+              sh.rows_per_stripe, nullptr, mask.data_ptr<uint8_t>());        \
+    else if (relu)                                                           \
       bn_norm_kernel<T, true, VEC><<<sh.grid, kBlock, 0, s>>>(               \
           tp<T>(x), tpm<T>(y), scale.data_ptr<float>(),                      \
           shift.data_ptr<float>(), sh.R, sh.C, sh.CT, sh.rows_per_stripe);   \
@@ -422,7 +458,7 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd_train_impl(
   else if (sh.vec) { DLX_BN_T(float, true) }
   else { DLX_BN_T(float, false) }
 #undef DLX_BN_T
-  return {y, save_mean, save_invstd};
+  return {y, save_mean, save_invstd, mask};
 }
 
 at::Tensor bn_fwd_eval_impl(const at::Tensor& x_in, const at::Tensor& w,
@@ -513,7 +549,8 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_bwd_impl(
 // dres = relu-masked dy in the same dx pass (one fewer elementwise pass
 // and two fewer kernels than bn + add_relu).
 
-std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd_train_res_impl(
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
+bn_fwd_train_res_impl(
     const at::Tensor& x_in, const at::Tensor& w, const at::Tensor& b,
     at::Tensor& rmean, at::Tensor& rvar, double momentum, double eps,
     const at::Tensor& res_in) {
@@ -530,6 +567,9 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd_train_res_impl(
   auto scale = at::empty({sh.C}, fopts);
   auto shift = at::empty({sh.C}, fopts);
   auto y = at::empty_like(x);
+  const int vw = bf ? 8 : 4;
+  auto mask = at::empty({sh.R * (sh.C / vw)},
+                        x.options().dtype(at::kByte));
 
 #define DLX_BN_RES(T)                                                        \
   {                                                                          \
@@ -542,14 +582,14 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd_train_res_impl(
         save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),          \
         scale.data_ptr<float>(), shift.data_ptr<float>(), sh.R, sh.C,        \
         sh.stripes, (float)momentum, (float)eps);                            \
-    bn_norm_kernel<T, true, true, true><<<sh.grid, kBlock, 0, s>>>(          \
+    bn_norm_kernel<T, true, true, true, true><<<sh.grid, kBlock, 0, s>>>(    \
         tp<T>(x), tpm<T>(y), scale.data_ptr<float>(),                        \
         shift.data_ptr<float>(), sh.R, sh.C, sh.CT, sh.rows_per_stripe,     \
-        tp<T>(res));                                                         \
+        tp<T>(res), mask.data_ptr<uint8_t>());                               \
   }
   if (bf) { DLX_BN_RES(uint16_t) } else { DLX_BN_RES(float) }
 #undef DLX_BN_RES
-  return {y, save_mean, save_invstd};
+  return {y, save_mean, save_invstd, mask};
 }
 
 at::Tensor bn_fwd_eval_res_impl(const at::Tensor& x_in, const at::Tensor& w,
@@ -623,6 +663,96 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_bwd_res_impl(
   }
   if (bf) { DLX_BN_BWDR(uint16_t) } else { DLX_BN_BWDR(float) }
 #undef DLX_BN_BWDR
+  return {dx, dgamma, dbeta, dres};
+}
+
+// ---- masked backward (relu mask from the fwd bit-mask; y never read) ---
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_bwd_m_impl(
+    const at::Tensor& dy_in, const at::Tensor& x_in, const at::Tensor& w,
+    const at::Tensor& save_mean, const at::Tensor& save_invstd,
+    const at::Tensor& mask) {
+  auto x = x_in.contiguous(at::MemoryFormat::ChannelsLast);
+  auto dy = dy_in.contiguous(at::MemoryFormat::ChannelsLast);
+  auto s = at::cuda::getCurrentCUDAStream();
+  const bool bf = x.scalar_type() == at::kBFloat16;
+  auto sh = bn_shape(x, bf ? 8 : 4);
+  TORCH_CHECK(sh.vec && mask.numel() == sh.R * (sh.C / (bf ? 8 : 4)),
+              "bn_bwd_m: mask/shape mismatch");
+  auto fopts = x.options().dtype(at::kFloat);
+  auto sums = at::empty({(long)sh.stripes * 2 * sh.C}, fopts);
+  auto dgamma = at::empty({sh.C}, fopts);
+  auto dbeta = at::empty({sh.C}, fopts);
+  auto cA = at::empty({sh.C}, fopts);
+  auto cB = at::empty({sh.C}, fopts);
+  auto cD = at::empty({sh.C}, fopts);
+  auto dx = at::empty_like(x);
+
+#define DLX_BN_BWDM(T)                                                       \
+  {                                                                          \
+    bn_reduce_bwd_kernel<T, true, true, true>                                \
+        <<<sh.grid, kBlock, sh.lds, s>>>(                                    \
+            tp<T>(dy), tp<T>(x), nullptr, save_mean.data_ptr<float>(),       \
+            save_invstd.data_ptr<float>(), sums.data_ptr<float>(), sh.R,     \
+            sh.C, sh.CT, sh.rows_per_stripe, mask.data_ptr<uint8_t>());      \
+    bn_finalize_bwd_kernel<<<sh.C, kBlock, 0, s>>>(                          \
+        sums.data_ptr<float>(), w.data_ptr<float>(),                         \
+        save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),          \
+        dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),                   \
+        cA.data_ptr<float>(), cB.data_ptr<float>(), cD.data_ptr<float>(),    \
+        sh.R, sh.C, sh.stripes);                                             \
+    bn_dx_kernel<T, true, true, false, true><<<sh.grid, kBlock, 0, s>>>(     \
+        tp<T>(dy), tp<T>(x), nullptr, cA.data_ptr<float>(),                  \
+        cB.data_ptr<float>(), cD.data_ptr<float>(), tpm<T>(dx), sh.R, sh.C,  \
+        sh.CT, sh.rows_per_stripe, nullptr, mask.data_ptr<uint8_t>());       \
+  }
+  if (bf) { DLX_BN_BWDM(uint16_t) } else { DLX_BN_BWDM(float) }
+#undef DLX_BN_BWDM
+  return {dx, dgamma, dbeta};
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_bwd_res_m_impl(
+    const at::Tensor& dy_in, const at::Tensor& x_in, const at::Tensor& w,
+    const at::Tensor& save_mean, const at::Tensor& save_invstd,
+    const at::Tensor& mask) {
+  auto x = x_in.contiguous(at::MemoryFormat::ChannelsLast);
+  auto dy = dy_in.contiguous(at::MemoryFormat::ChannelsLast);
+  auto s = at::cuda::getCurrentCUDAStream();
+  const bool bf = x.scalar_type() == at::kBFloat16;
+  auto sh = bn_shape(x, bf ? 8 : 4);
+  TORCH_CHECK(sh.vec && mask.numel() == sh.R * (sh.C / (bf ? 8 : 4)),
+              "bn_bwd_res_m: mask/shape mismatch");
+  auto fopts = x.options().dtype(at::kFloat);
+  auto sums = at::empty({(long)sh.stripes * 2 * sh.C}, fopts);
+  auto dgamma = at::empty({sh.C}, fopts);
+  auto dbeta = at::empty({sh.C}, fopts);
+  auto cA = at::empty({sh.C}, fopts);
+  auto cB = at::empty({sh.C}, fopts);
+  auto cD = at::empty({sh.C}, fopts);
+  auto dx = at::empty_like(x);
+  auto dres = at::empty_like(x);
+
+#define DLX_BN_BWDRM(T)                                                      \
+  {                                                                          \
+    bn_reduce_bwd_kernel<T, true, true, true>                                \
+        <<<sh.grid, kBlock, sh.lds, s>>>(                                    \
+            tp<T>(dy), tp<T>(x), nullptr, save_mean.data_ptr<float>(),       \
+            save_invstd.data_ptr<float>(), sums.data_ptr<float>(), sh.R,     \
+            sh.C, sh.CT, sh.rows_per_stripe, mask.data_ptr<uint8_t>());      \
+    bn_finalize_bwd_kernel<<<sh.C, kBlock, 0, s>>>(                          \
+        sums.data_ptr<float>(), w.data_ptr<float>(),                         \
+        save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),          \
+        dgamma.data_ptr<float>(), dbeta.data_ptr<float>(),                   \
+        cA.data_ptr<float>(), cB.data_ptr<float>(), cD.data_ptr<float>(),    \
+        sh.R, sh.C, sh.stripes);                                             \
+    bn_dx_kernel<T, true, true, true, true><<<sh.grid, kBlock, 0, s>>>(      \
+        tp<T>(dy), tp<T>(x), nullptr, cA.data_ptr<float>(),                  \
+        cB.data_ptr<float>(), cD.data_ptr<float>(), tpm<T>(dx), sh.R, sh.C,  \
+        sh.CT, sh.rows_per_stripe, tpm<T>(dres),                             \
+        mask.data_ptr<uint8_t>());                                           \
+  }
+  if (bf) { DLX_BN_BWDRM(uint16_t) } else { DLX_BN_BWDRM(float) }
+#undef DLX_BN_BWDRM
   return {dx, dgamma, dbeta, dres};
 }
 

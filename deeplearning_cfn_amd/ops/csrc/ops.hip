@@ -38,14 +38,24 @@ at::Tensor gemm_nt_impl(const at::Tensor& A, const at::Tensor& B,
                         bool out_bf16);
 
 // batchnorm.hip
-std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd_train_impl(
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
+bn_fwd_train_impl(
     const at::Tensor& x, const at::Tensor& w, const at::Tensor& b,
     at::Tensor& rmean, at::Tensor& rvar, double momentum, double eps,
     bool relu);
 at::Tensor bn_fwd_eval_impl(const at::Tensor& x, const at::Tensor& w,
                             const at::Tensor& b, const at::Tensor& rmean,
                             const at::Tensor& rvar, double eps, bool relu);
-std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_fwd_train_res_impl(
+std::tuple<at::Tensor, at::Tensor, at::Tensor> bn_bwd_m_impl(
+    const at::Tensor& dy, const at::Tensor& x, const at::Tensor& w,
+    const at::Tensor& save_mean, const at::Tensor& save_invstd,
+    const at::Tensor& mask);
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_bwd_res_m_impl(
+    const at::Tensor& dy, const at::Tensor& x, const at::Tensor& w,
+    const at::Tensor& save_mean, const at::Tensor& save_invstd,
+    const at::Tensor& mask);
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
+bn_fwd_train_res_impl(
     const at::Tensor& x, const at::Tensor& w, const at::Tensor& b,
     at::Tensor& rmean, at::Tensor& rvar, double momentum, double eps,
     const at::Tensor& res);
@@ -238,10 +248,12 @@ TORCH_LIBRARY(dlx_hip, m) {
   m.def("adamw_flat_master(Tensor(a!) p, Tensor g, Tensor(b!) m, Tensor(c!) v, Tensor(d!) master, float lr, float b1, float b2, float eps, float wd, float bc1, float bc2) -> ()");
   m.def("softmax_ce_fwd(Tensor logits, Tensor target) -> (Tensor, Tensor)");
   m.def("softmax_ce_bwd(Tensor probs, Tensor target, Tensor grad, bool out_bf16) -> Tensor");
-  m.def("bn_fwd_train(Tensor x, Tensor w, Tensor b, Tensor(a!) rmean, Tensor(b!) rvar, float momentum, float eps, bool relu) -> (Tensor, Tensor, Tensor)");
+  m.def("bn_fwd_train(Tensor x, Tensor w, Tensor b, Tensor(a!) rmean, Tensor(b!) rvar, float momentum, float eps, bool relu) -> (Tensor, Tensor, Tensor, Tensor)");
   m.def("bn_fwd_eval(Tensor x, Tensor w, Tensor b, Tensor rmean, Tensor rvar, float eps, bool relu) -> Tensor");
   m.def("bn_bwd(Tensor dy, Tensor x, Tensor w, Tensor save_mean, Tensor save_invstd, Tensor y, bool relu) -> (Tensor, Tensor, Tensor)");
-  m.def("bn_fwd_train_res(Tensor x, Tensor w, Tensor b, Tensor(a!) rmean, Tensor(b!) rvar, float momentum, float eps, Tensor res) -> (Tensor, Tensor, Tensor)");
+  m.def("bn_fwd_train_res(Tensor x, Tensor w, Tensor b, Tensor(a!) rmean, Tensor(b!) rvar, float momentum, float eps, Tensor res) -> (Tensor, Tensor, Tensor, Tensor)");
+  m.def("bn_bwd_m(Tensor dy, Tensor x, Tensor w, Tensor save_mean, Tensor save_invstd, Tensor mask) -> (Tensor, Tensor, Tensor)");
+  m.def("bn_bwd_res_m(Tensor dy, Tensor x, Tensor w, Tensor save_mean, Tensor save_invstd, Tensor mask) -> (Tensor, Tensor, Tensor, Tensor)");
   m.def("bn_fwd_eval_res(Tensor x, Tensor w, Tensor b, Tensor rmean, Tensor rvar, float eps, Tensor res) -> Tensor");
   m.def("bn_bwd_res(Tensor dy, Tensor x, Tensor w, Tensor save_mean, Tensor save_invstd, Tensor y) -> (Tensor, Tensor, Tensor, Tensor)");
   m.def("gemm_nt(Tensor a, Tensor b) -> Tensor");
@@ -279,6 +291,8 @@ TORCH_LIBRARY_IMPL(dlx_hip, CUDA, m) {
   m.impl("bn_fwd_train_res", dlx::bn_fwd_train_res_impl);
   m.impl("bn_fwd_eval_res", dlx::bn_fwd_eval_res_impl);
   m.impl("bn_bwd_res", dlx::bn_bwd_res_impl);
+  m.impl("bn_bwd_m", dlx::bn_bwd_m_impl);
+  m.impl("bn_bwd_res_m", dlx::bn_bwd_res_m_impl);
   m.impl("gemm_nt", dlx::gemm_nt);
   m.impl("gemm_nn", dlx::gemm_nn);
   m.impl("gemm_tn", dlx::gemm_tn);

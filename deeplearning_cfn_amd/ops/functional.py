@@ -79,17 +79,28 @@ class _BNAct(torch.autograd.Function):
     def forward(ctx, x, weight, bias, running_mean, running_var,
                 momentum: float, eps: float, relu_: bool):
         x = _to_nhwc(x)
-        y, save_mean, save_invstd = native().bn_fwd_train(
+        y, save_mean, save_invstd, mask = native().bn_fwd_train(
             x, weight, bias, running_mean, running_var, momentum, eps, relu_)
-        ctx.save_for_backward(x, weight, save_mean, save_invstd, y)
+        # with a relu bit-mask, backward never re-reads y (2 fewer passes)
+        if mask.numel():
+            ctx.save_for_backward(x, weight, save_mean, save_invstd, mask)
+            ctx.masked = True
+        else:
+            ctx.save_for_backward(x, weight, save_mean, save_invstd, y)
+            ctx.masked = False
         ctx.relu = relu_
         return y
 
     @staticmethod
     def backward(ctx, grad: torch.Tensor):
-        x, weight, save_mean, save_invstd, y = ctx.saved_tensors
-        dx, dweight, dbias = native().bn_bwd(
-            _to_nhwc(grad), x, weight, save_mean, save_invstd, y, ctx.relu)
+        x, weight, save_mean, save_invstd, aux = ctx.saved_tensors
+        if ctx.masked:
+            dx, dweight, dbias = native().bn_bwd_m(
+                _to_nhwc(grad), x, weight, save_mean, save_invstd, aux)
+        else:
+            dx, dweight, dbias = native().bn_bwd(
+                _to_nhwc(grad), x, weight, save_mean, save_invstd, aux,
+                ctx.relu)
         return dx, dweight, dbias, None, None, None, None, None
 
 
@@ -122,17 +133,17 @@ class _BNAddReLU(torch.autograd.Function):
     def forward(ctx, x, residual, weight, bias, running_mean, running_var,
                 momentum: float, eps: float):
         x = _to_nhwc(x)
-        y, save_mean, save_invstd = native().bn_fwd_train_res(
+        y, save_mean, save_invstd, mask = native().bn_fwd_train_res(
             x, weight, bias, running_mean, running_var, momentum, eps,
             _to_nhwc(residual))
-        ctx.save_for_backward(x, weight, save_mean, save_invstd, y)
+        ctx.save_for_backward(x, weight, save_mean, save_invstd, mask)
         return y
 
     @staticmethod
     def backward(ctx, grad: torch.Tensor):
-        x, weight, save_mean, save_invstd, y = ctx.saved_tensors
-        dx, dweight, dbias, dres = native().bn_bwd_res(
-            _to_nhwc(grad), x, weight, save_mean, save_invstd, y)
+        x, weight, save_mean, save_invstd, mask = ctx.saved_tensors
+        dx, dweight, dbias, dres = native().bn_bwd_res_m(
+            _to_nhwc(grad), x, weight, save_mean, save_invstd, mask)
         return dx, dres, dweight, dbias, None, None, None, None
 
 

@@ -60,7 +60,7 @@ def test_bn_fwd_train_parity(dtype, relu, C):
     rvar = torch.ones(C, device="cuda")
     rmean_ref, rvar_ref = rmean.clone(), rvar.clone()
 
-    y, mean, invstd = n.bn_fwd_train(x, w, b, rmean, rvar, 0.1, 1e-5, relu)
+    y, mean, invstd, _mask = n.bn_fwd_train(x, w, b, rmean, rvar, 0.1, 1e-5, relu)
     ref = torch.nn.functional.batch_norm(
         x.float(), rmean_ref, rvar_ref, w, b, True, 0.1, 1e-5)
     if relu:
@@ -89,9 +89,15 @@ def test_bn_bwd_parity(relu):
     dy = _rand((4, C, 6, 6), torch.float32, cl=True, seed=6)
     ref.backward(dy)
 
-    y, mean, invstd = n.bn_fwd_train(xf, w.detach(), b.detach(), rm, rv,
+    y, mean, invstd, mask = n.bn_fwd_train(xf, w.detach(), b.detach(), rm, rv,
                                      0.1, 1e-5, relu)
     dx, dw, db = n.bn_bwd(dy, xf, w.detach(), mean, invstd, y, relu)
+    if mask.numel():
+        # masked backward must agree with the y-based one
+        dx2, dw2, db2 = n.bn_bwd_m(dy, xf, w.detach(), mean, invstd, mask)
+        torch.testing.assert_close(dx2, dx, rtol=0, atol=0)
+        torch.testing.assert_close(dw2, dw, rtol=0, atol=0)
+        torch.testing.assert_close(db2, db, rtol=0, atol=0)
     torch.testing.assert_close(dx.float(), x.grad, rtol=1e-3, atol=1e-4)
     torch.testing.assert_close(dw, w.grad, rtol=1e-3, atol=1e-3)
     torch.testing.assert_close(db, b.grad, rtol=1e-3, atol=1e-3)
