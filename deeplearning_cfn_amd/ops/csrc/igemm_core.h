@@ -12,10 +12,11 @@
 //  * fragment ds_read_b128 is conflict-free: each 16-lane group reads 16
 //    CONSECUTIVE 16-B slots (one full 256-B bank row).
 //
-// Geometries: 128x128 (4 waves, 2x2) for narrow problems; 256x256
-// (8 waves, 2x4, 128 KiB LDS, 1 block/CU) for the big ones.
-// Double-buffered, stage(t+1) issued before compute(t), one
-// vmcnt(0)+barrier per K-tile (guide §5.5 T3 minimum 2-phase).
+// Geometries: 256x256 8-phase counted-vmcnt pipeline (8 waves, 128 KiB
+// LDS) for big problems; 256x64 (narrow N) and 128x128 2-phase tiles
+// otherwise.  Output rows go through an OM mapper (identity, or the
+// stride-2 dgrad parity-class pixel scatter).  1x1/s1 convs are routed
+// with plain RowGathers (no im2col address math) by the conv host code.
 #pragma once
 
 #include <cstdlib>
