@@ -454,7 +454,13 @@ template <class AG, class BG, bool OUT_BF16, class OM>
 inline void igemm_launch_om(const AG& ag, const BG& bg, void* cptr,
                             const uint16_t* zp, long M, long Nc, long K,
                             hipStream_t s, const OM& om) {
-  if (M >= 256 && Nc >= 192 && K >= 128) {
+  // 8ph needs a long K-loop to fill its counted-vmcnt pipeline (its one
+  // 128-KiB block per CU cannot hide the prologue/drain otherwise)
+  static const long mink8 = []() {
+    const char* e = getenv("DLX_8PH_MINK");
+    return e ? atol(e) : 128;
+  }();
+  if (M >= 256 && Nc >= 192 && K >= mink8) {
     const int gm = (int)((M + 255) / 256), gn = (int)((Nc + 255) / 256);
     igemm_kernel_8ph<AG, BG, OUT_BF16, OM>
         <<<gm * gn, 512, 0, s>>>(ag, bg, cptr, zp, M, Nc, K, gn, om);
