@@ -458,7 +458,7 @@ inline void igemm_launch_om(const AG& ag, const BG& bg, void* cptr,
   // 128-KiB block per CU cannot hide the prologue/drain otherwise)
   static const long mink8 = []() {
     const char* e = getenv("DLX_8PH_MINK");
-    return e ? atol(e) : 128;
+    return e ? atol(e) : 384;
   }();
   if (M >= 256 && Nc >= 192 && K >= mink8) {
     const int gm = (int)((M + 255) / 256), gn = (int)((Nc + 255) / 256);
