@@ -78,17 +78,21 @@ def main() -> int:
     # Pre-generate a rotating set of on-device batches: the metric is the
     # training step (the reference's synthetic-data benchmark mode skips
     # input pipelines the same way); no work is skipped inside the step.
-    g = torch.Generator().manual_seed(1000 + ctx.rank)
+    # Generated on-device: CPU randn of 4x 1.2 GB (b2048) costs tens of
+    # seconds of startup per invocation for identical synthetic content.
+    g = (torch.Generator(device=ctx.device).manual_seed(1000 + ctx.rank)
+         if cuda else torch.Generator().manual_seed(1000 + ctx.rank))
     nbatch = 4
     xs, ys = [], []
     for _ in range(nbatch):
-        x = torch.randn(args.batch_size, 3, 224, 224, generator=g).to(dtype)
-        y = torch.randint(0, 1000, (args.batch_size,), generator=g)
-        x = x.to(ctx.device)
+        x = torch.randn(args.batch_size, 3, 224, 224, generator=g,
+                        device=ctx.device, dtype=torch.float32).to(dtype)
+        y = torch.randint(0, 1000, (args.batch_size,), generator=g,
+                          device=ctx.device)
         if cuda:
             x = x.contiguous(memory_format=torch.channels_last)
         xs.append(x)
-        ys.append(y.to(ctx.device))
+        ys.append(y)
 
     def raw_step(x, y):
         engine.zero_grad()
