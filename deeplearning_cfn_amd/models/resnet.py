@@ -56,6 +56,20 @@ class GlobalAvgPool(nn.Module):
         return F_amd.global_avg_pool(x)
 
 
+class ConvBN(nn.Module):
+    """Downsample path: 1x1 conv + BN (no relu), fused on the native path."""
+
+    def __init__(self, cin: int, cout: int, stride: int):
+        super().__init__()
+        self.conv = conv1x1(cin, cout, stride)
+        self.bn = nn.BatchNorm2d(cout)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return F_amd.conv_bn_act(x, self.conv.weight, self.bn,
+                                 self.conv.stride[0], 0, relu=False,
+                                 training=self.training)
+
+
 class BNReLU(nn.Module):
     """BatchNorm2d + ReLU, fused on the native path (ops/bn_relu kernel)."""
 
@@ -83,10 +97,12 @@ class BasicBlock(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         identity = x if self.downsample is None else self.downsample(x)
-        out = self.bn1(self.conv1(x))
-        out = self.conv2(out)
-        return F_amd.batch_norm_add_relu(out, self.bn2.bn, identity,
-                                         self.training)
+        out = F_amd.conv_bn_act(x, self.conv1.weight, self.bn1.bn,
+                                self.conv1.stride[0], self.conv1.padding[0],
+                                relu=True, training=self.training)
+        return F_amd.conv_bn_act(out, self.conv2.weight, self.bn2.bn,
+                                 self.conv2.stride[0], self.conv2.padding[0],
+                                 residual=identity, training=self.training)
 
 
 class Bottleneck(nn.Module):
@@ -105,11 +121,14 @@ class Bottleneck(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         identity = x if self.downsample is None else self.downsample(x)
-        out = self.bn1(self.conv1(x))
-        out = self.bn2(self.conv2(out))
-        out = self.conv3(out)
-        return F_amd.batch_norm_add_relu(out, self.bn3.bn, identity,
-                                         self.training)
+        out = F_amd.conv_bn_act(x, self.conv1.weight, self.bn1.bn,
+                                1, 0, relu=True, training=self.training)
+        out = F_amd.conv_bn_act(out, self.conv2.weight, self.bn2.bn,
+                                self.conv2.stride[0], self.conv2.padding[0],
+                                relu=True, training=self.training)
+        return F_amd.conv_bn_act(out, self.conv3.weight, self.bn3.bn,
+                                 1, 0, residual=identity,
+                                 training=self.training)
 
 
 class ResNet(nn.Module):
@@ -155,10 +174,8 @@ class ResNet(nn.Module):
                     stride: int) -> nn.Sequential:
         downsample = None
         if stride != 1 or self.inplanes != planes * block.expansion:
-            downsample = nn.Sequential(
-                conv1x1(self.inplanes, planes * block.expansion, stride),
-                BNReLU(planes * block.expansion, relu=False),
-            )
+            downsample = ConvBN(self.inplanes, planes * block.expansion,
+                                stride)
         layers = [block(self.inplanes, planes, stride, downsample)]
         self.inplanes = planes * block.expansion
         for _ in range(1, blocks):

@@ -756,4 +756,94 @@ std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor> bn_bwd_res_m_impl(
   return {dx, dgamma, dbeta, dres};
 }
 
+// ---- presum forward: channel sums come from the conv epilogue ---------
+// (conv2d_fwd_stats slab [stripes][2][C]); only finalize + normalize run.
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
+bn_fwd_train_presum_impl(
+    const at::Tensor& x_in, const at::Tensor& w, const at::Tensor& b,
+    at::Tensor& rmean, at::Tensor& rvar, double momentum, double eps,
+    bool relu, const at::Tensor& sums) {
+  auto x = x_in.contiguous(at::MemoryFormat::ChannelsLast);
+  auto s = at::cuda::getCurrentCUDAStream();
+  const bool bf = x.scalar_type() == at::kBFloat16;
+  auto sh = bn_shape(x, bf ? 8 : 4);
+  TORCH_CHECK(sh.vec, "bn presum: C % vector width != 0");
+  const int nstripes = (int)(sums.numel() / (2 * sh.C));
+  TORCH_CHECK((long)nstripes * 2 * sh.C == sums.numel(),
+              "bn presum: sums size mismatch");
+  auto fopts = x.options().dtype(at::kFloat);
+  auto save_mean = at::empty({sh.C}, fopts);
+  auto save_invstd = at::empty({sh.C}, fopts);
+  auto scale = at::empty({sh.C}, fopts);
+  auto shift = at::empty({sh.C}, fopts);
+  auto y = at::empty_like(x);
+  const int vw = bf ? 8 : 4;
+  auto mask = relu ? at::empty({sh.R * (sh.C / vw)},
+                               x.options().dtype(at::kByte))
+                   : at::empty({0}, x.options().dtype(at::kByte));
+
+#define DLX_BN_PRE(T)                                                        \
+  {                                                                          \
+    bn_finalize_fwd_kernel<<<sh.C, kBlock, 0, s>>>(                          \
+        sums.data_ptr<float>(), w.data_ptr<float>(), b.data_ptr<float>(),    \
+        rmean.data_ptr<float>(), rvar.data_ptr<float>(),                     \
+        save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),          \
+        scale.data_ptr<float>(), shift.data_ptr<float>(), sh.R, sh.C,        \
+        nstripes, (float)momentum, (float)eps);                              \
+    if (relu)                                                                \
+      bn_norm_kernel<T, true, true, false, true>                             \
+          <<<sh.grid, kBlock, 0, s>>>(                                       \
+              tp<T>(x), tpm<T>(y), scale.data_ptr<float>(),                  \
+              shift.data_ptr<float>(), sh.R, sh.C, sh.CT,                    \
+              sh.rows_per_stripe, nullptr, mask.data_ptr<uint8_t>());        \
+    else                                                                     \
+      bn_norm_kernel<T, false, true><<<sh.grid, kBlock, 0, s>>>(             \
+          tp<T>(x), tpm<T>(y), scale.data_ptr<float>(),                      \
+          shift.data_ptr<float>(), sh.R, sh.C, sh.CT, sh.rows_per_stripe);   \
+  }
+  if (bf) { DLX_BN_PRE(uint16_t) } else { DLX_BN_PRE(float) }
+#undef DLX_BN_PRE
+  return {y, save_mean, save_invstd, mask};
+}
+
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
+bn_fwd_train_res_presum_impl(
+    const at::Tensor& x_in, const at::Tensor& w, const at::Tensor& b,
+    at::Tensor& rmean, at::Tensor& rvar, double momentum, double eps,
+    const at::Tensor& res_in, const at::Tensor& sums) {
+  auto x = x_in.contiguous(at::MemoryFormat::ChannelsLast);
+  auto res = res_in.contiguous(at::MemoryFormat::ChannelsLast);
+  auto s = at::cuda::getCurrentCUDAStream();
+  const bool bf = x.scalar_type() == at::kBFloat16;
+  auto sh = bn_shape(x, bf ? 8 : 4);
+  TORCH_CHECK(sh.vec, "bn presum: C % vector width != 0");
+  const int nstripes = (int)(sums.numel() / (2 * sh.C));
+  auto fopts = x.options().dtype(at::kFloat);
+  auto save_mean = at::empty({sh.C}, fopts);
+  auto save_invstd = at::empty({sh.C}, fopts);
+  auto scale = at::empty({sh.C}, fopts);
+  auto shift = at::empty({sh.C}, fopts);
+  auto y = at::empty_like(x);
+  const int vw = bf ? 8 : 4;
+  auto mask = at::empty({sh.R * (sh.C / vw)}, x.options().dtype(at::kByte));
+
+#define DLX_BN_PRER(T)                                                       \
+  {                                                                          \
+    bn_finalize_fwd_kernel<<<sh.C, kBlock, 0, s>>>(                          \
+        sums.data_ptr<float>(), w.data_ptr<float>(), b.data_ptr<float>(),    \
+        rmean.data_ptr<float>(), rvar.data_ptr<float>(),                     \
+        save_mean.data_ptr<float>(), save_invstd.data_ptr<float>(),          \
+        scale.data_ptr<float>(), shift.data_ptr<float>(), sh.R, sh.C,        \
+        nstripes, (float)momentum, (float)eps);                              \
+    bn_norm_kernel<T, true, true, true, true><<<sh.grid, kBlock, 0, s>>>(    \
+        tp<T>(x), tpm<T>(y), scale.data_ptr<float>(),                        \
+        shift.data_ptr<float>(), sh.R, sh.C, sh.CT, sh.rows_per_stripe,     \
+        tp<T>(res), mask.data_ptr<uint8_t>());                               \
+  }
+  if (bf) { DLX_BN_PRER(uint16_t) } else { DLX_BN_PRER(float) }
+#undef DLX_BN_PRER
+  return {y, save_mean, save_invstd, mask};
+}
+
 }  // namespace dlx

@@ -66,6 +66,48 @@ at::Tensor conv2d_fwd_impl(const at::Tensor& x_in, const at::Tensor& w_in,
   return y;
 }
 
+// conv fwd + per-block channel stats: returns (y, sums[gm][2][Co]) — the
+// BN reduce pass collapses into this epilogue (bn_fwd_train_presum).
+std::tuple<at::Tensor, at::Tensor> conv2d_fwd_stats_impl(
+    const at::Tensor& x_in, const at::Tensor& w_in, int64_t stride,
+    int64_t pad) {
+  auto x = x_in.contiguous(at::MemoryFormat::ChannelsLast);
+  auto w = w_in.contiguous(at::MemoryFormat::ChannelsLast);
+  TORCH_CHECK(x.scalar_type() == at::kBFloat16, "conv: bf16 only");
+  ConvShape p;
+  p.N = x.size(0); p.Ci = x.size(1); p.H = x.size(2); p.W = x.size(3);
+  p.Co = w.size(0); p.Kh = w.size(2); p.Kw = w.size(3);
+  p.stride = (int)stride; p.pad = (int)pad;
+  p.Ho = (p.H + 2 * p.pad - p.Kh) / p.stride + 1;
+  p.Wo = (p.W + 2 * p.pad - p.Kw) / p.stride + 1;
+  p.K = (long)p.Kh * p.Kw * p.Ci;
+  p.M = (long)p.N * p.Ho * p.Wo;
+  p.finalize();
+  TORCH_CHECK(p.Ci % 8 == 0, "conv fwd: Cin % 8 != 0 (use library path)");
+
+  auto y = at::empty({p.N, p.Co, p.Ho, p.Wo},
+                     x.options().memory_format(at::MemoryFormat::ChannelsLast));
+  const long gm = igemm_gm(p.M, p.Co, p.K);
+  auto sums = at::empty({gm * 2 * p.Co}, x.options().dtype(at::kFloat));
+  auto zp = zero_page_for(x);
+  auto s = at::cuda::getCurrentCUDAStream();
+  RowGather bg{reinterpret_cast<const uint16_t*>(w.data_ptr()), p.Co, p.K};
+  if (p.Kh == 1 && p.Kw == 1 && p.stride == 1 && p.pad == 0) {
+    RowGather ag{reinterpret_cast<const uint16_t*>(x.data_ptr()), p.M, p.K};
+    igemm_launch_stats<RowGather, RowGather>(
+        ag, bg, y.data_ptr(),
+        reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, p.Co, p.K, s,
+        sums.data_ptr<float>());
+  } else {
+    ConvFwdGather ag{reinterpret_cast<const uint16_t*>(x.data_ptr()), p};
+    igemm_launch_stats<ConvFwdGather, RowGather>(
+        ag, bg, y.data_ptr(),
+        reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, p.Co, p.K, s,
+        sums.data_ptr<float>());
+  }
+  return {y, sums};
+}
+
 // g: [N,Co,Ho,Wo] CL; w: [Co,Ci,Kh,Kw]; returns dx [N,Ci,H,W] CL
 at::Tensor conv2d_dgrad_impl(const at::Tensor& g_in, const at::Tensor& w_in,
                              int64_t H, int64_t W, int64_t stride,

@@ -179,12 +179,60 @@ struct Stride2Out {
 
 // ---- the kernel --------------------------------------------------------
 
+// Deterministic per-col (channel) sum / sum-of-squares of this block's
+// output tile, written to stats[bm][2][Nc] — lets BatchNorm skip its
+// whole-tensor reduce pass (the conv epilogue already holds y in regs).
+// sred: LDS scratch (reused staging memory, >= 2*BN*WM floats).
+template <int BM, int BN, int WM, int WN, int MR, int NR>
+DLX_DEV void igemm_stats_epilogue(const f32x4 (*acc)[NR], float* sred,
+                                  float* __restrict__ stats, long bm,
+                                  long n0, long Nc, int wr, int wc, int l4,
+                                  int l15, int tid) {
+#pragma unroll
+  for (int ni = 0; ni < NR; ++ni) {
+    float s1 = 0.f, s2 = 0.f;
+#pragma unroll
+    for (int mi = 0; mi < MR; ++mi)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const float v = acc[mi][ni][r];
+        s1 += v;
+        s2 += v * v;
+      }
+    // sum the 4 row-groups (l4) of this wave: lanes l<16 end up valid
+    s1 += __shfl_down(s1, 16, kWave);
+    s1 += __shfl_down(s1, 32, kWave);
+    s2 += __shfl_down(s2, 16, kWave);
+    s2 += __shfl_down(s2, 32, kWave);
+    if (l4 == 0) {
+      const int cl = wc * (BN / WN) + ni * 16 + l15;  // block-local col
+      sred[(0 * BN + cl) * WM + wr] = s1;
+      sred[(1 * BN + cl) * WM + wr] = s2;
+    }
+  }
+  __syncthreads();
+  // threads 0..BN-1: fixed-order sum over the WM row-waves (deterministic)
+  for (int cl = tid; cl < BN; cl += WM * WN * 64) {
+    if (n0 + cl >= Nc) continue;
+    float t1 = 0.f, t2 = 0.f;
+#pragma unroll
+    for (int w = 0; w < WM; ++w) {
+      t1 += sred[(0 * BN + cl) * WM + w];
+      t2 += sred[(1 * BN + cl) * WM + w];
+    }
+    stats[(bm * 2 + 0) * Nc + n0 + cl] = t1;
+    stats[(bm * 2 + 1) * Nc + n0 + cl] = t2;
+  }
+}
+
 template <int BM, int BN, int BK, int WM, int WN,
-          class AG, class BG, bool OUT_BF16, class OM = IdentityOut>
+          class AG, class BG, bool OUT_BF16, class OM = IdentityOut,
+          bool STATS = false>
 __global__ __launch_bounds__(WM* WN * 64) void igemm_kernel(
     AG ag, BG bg, void* __restrict__ Cout_,
     const uint16_t* __restrict__ zero_page,
-    long M, long Nc, long K, int grid_n, OM om) {
+    long M, long Nc, long K, int grid_n, OM om,
+    float* __restrict__ stats = nullptr) {
   constexpr int THREADS = WM * WN * 64;
   constexpr int SLOTS_A = (BK / 8) * BM;
   constexpr int SLOTS_B = (BK / 8) * BN;
@@ -295,6 +343,12 @@ __global__ __launch_bounds__(WM* WN * 64) void igemm_kernel(
       }
     }
   }
+  if (STATS) {
+    // staging LDS is dead after the loop's final barrier — reuse it
+    igemm_stats_epilogue<BM, BN, WM, WN, MR, NR>(
+        acc, reinterpret_cast<float*>(lsA), stats, bm, n0, Nc, wr, wc, l4,
+        l15, threadIdx.x);
+  }
 }
 
 // ---------------------------------------------------------------------
@@ -311,11 +365,13 @@ __global__ __launch_bounds__(WM* WN * 64) void igemm_kernel(
 //    creates the wave role diversity it needs).
 // ---------------------------------------------------------------------
 
-template <class AG, class BG, bool OUT_BF16, class OM = IdentityOut>
+template <class AG, class BG, bool OUT_BF16, class OM = IdentityOut,
+          bool STATS = false>
 __global__ __launch_bounds__(512) void igemm_kernel_8ph(
     AG ag, BG bg, void* __restrict__ Cout_,
     const uint16_t* __restrict__ zero_page,
-    long M, long Nc, long K, int grid_n, OM om) {
+    long M, long Nc, long K, int grid_n, OM om,
+    float* __restrict__ stats = nullptr) {
   constexpr int BM = 256, BN = 256, BK = 64;
   constexpr int THREADS = 512;
   constexpr int SLOTS = (BK / 8) * BM;          // 2048 16-B slots/operand
@@ -446,6 +502,12 @@ __global__ __launch_bounds__(512) void igemm_kernel_8ph(
       }
     }
   }
+  if (STATS) {
+    __syncthreads();  // last phase has no trailing barrier; LDS now dead
+    igemm_stats_epilogue<256, 256, 2, 4, MR, NR>(
+        acc, reinterpret_cast<float*>(lds), stats, bm, n0, Nc, wr, wc, l4,
+        l15, threadIdx.x);
+  }
 }
 
 // launch helper: picks the 8-phase 256^2 pipeline for big problems,
@@ -493,6 +555,54 @@ inline void igemm_launch(const AG& ag, const BG& bg, void* cptr,
                          hipStream_t s) {
   igemm_launch_om<AG, BG, OUT_BF16, IdentityOut>(ag, bg, cptr, zp, M, Nc, K,
                                                  s, IdentityOut{Nc});
+}
+
+// geometry the launcher will pick (the stats slab is [gm][2][Nc], so the
+// host must know gm before allocating)
+inline long igemm_gm(long M, long Nc, long K) {
+  static const long mink8 = []() {
+    const char* e = getenv("DLX_8PH_MINK");
+    return e ? atol(e) : 384;
+  }();
+  if (M >= 256 && Nc >= 192 && K >= mink8) return (M + 255) / 256;
+  if (Nc <= 96 && M >= 1024) return (M + 255) / 256;
+  return (M + 127) / 128;
+}
+
+// conv-fwd launch with the per-block channel-stats epilogue (stats
+// slab [igemm_gm][2][Nc] fp32; reduce with the BN finalize kernel)
+template <class AG, class BG>
+inline void igemm_launch_stats(const AG& ag, const BG& bg, void* cptr,
+                               const uint16_t* zp, long M, long Nc, long K,
+                               hipStream_t s, float* stats) {
+  static const long mink8 = []() {
+    const char* e = getenv("DLX_8PH_MINK");
+    return e ? atol(e) : 384;
+  }();
+  const IdentityOut om{Nc};
+  if (M >= 256 && Nc >= 192 && K >= mink8) {
+    const int gm = (int)((M + 255) / 256), gn = (int)((Nc + 255) / 256);
+    igemm_kernel_8ph<AG, BG, true, IdentityOut, true>
+        <<<gm * gn, 512, 0, s>>>(ag, bg, cptr, zp, M, Nc, K, gn, om, stats);
+  } else if (Nc <= 96 && M >= 1024) {
+    const int gm = (int)((M + 255) / 256), gn = (int)((Nc + 63) / 64);
+    igemm_kernel<256, 64, 64, 4, 1, AG, BG, true, IdentityOut, true>
+        <<<gm * gn, 256, 0, s>>>(ag, bg, cptr, zp, M, Nc, K, gn, om, stats);
+  } else {
+    static const bool w8 = []() {
+      const char* e = getenv("DLX_IGEMM_W8");
+      return !(e && e[0] == '0');
+    }();
+    const int gm = (int)((M + 127) / 128), gn = (int)((Nc + 127) / 128);
+    if (w8)
+      igemm_kernel<128, 128, 64, 2, 4, AG, BG, true, IdentityOut, true>
+          <<<gm * gn, 512, 0, s>>>(ag, bg, cptr, zp, M, Nc, K, gn, om,
+                                   stats);
+    else
+      igemm_kernel<128, 128, 64, 2, 2, AG, BG, true, IdentityOut, true>
+          <<<gm * gn, 256, 0, s>>>(ag, bg, cptr, zp, M, Nc, K, gn, om,
+                                   stats);
+  }
 }
 
 }  // namespace dlx

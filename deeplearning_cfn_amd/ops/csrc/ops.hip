@@ -108,6 +108,20 @@ at::Tensor gelu_bwd_impl(const at::Tensor& g, const at::Tensor& x);
 // conv_wrw.hip — reduce-major GEMM (Linear wgrad)
 at::Tensor gemm_tn_mred_impl(const at::Tensor& a, const at::Tensor& b);
 
+// conv fwd with channel-stats epilogue + BN presum forward (fusion seam)
+std::tuple<at::Tensor, at::Tensor> conv2d_fwd_stats_impl(
+    const at::Tensor& x, const at::Tensor& w, int64_t stride, int64_t pad);
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
+bn_fwd_train_presum_impl(
+    const at::Tensor& x, const at::Tensor& w, const at::Tensor& b,
+    at::Tensor& rmean, at::Tensor& rvar, double momentum, double eps,
+    bool relu, const at::Tensor& sums);
+std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
+bn_fwd_train_res_presum_impl(
+    const at::Tensor& x, const at::Tensor& w, const at::Tensor& b,
+    at::Tensor& rmean, at::Tensor& rvar, double momentum, double eps,
+    const at::Tensor& res, const at::Tensor& sums);
+
 namespace {
 
 hipStream_t cur_stream() { return at::cuda::getCurrentCUDAStream(); }
@@ -269,6 +283,9 @@ TORCH_LIBRARY(dlx_hip, m) {
   m.def("conv2d_fwd(Tensor x, Tensor w, int stride, int pad) -> Tensor");
   m.def("conv2d_dgrad(Tensor g, Tensor w, int H, int W, int stride, int pad) -> Tensor");
   m.def("conv2d_wrw(Tensor g, Tensor x, int Kh, int Kw, int stride, int pad) -> Tensor");
+  m.def("conv2d_fwd_stats(Tensor x, Tensor w, int stride, int pad) -> (Tensor, Tensor)");
+  m.def("bn_fwd_train_presum(Tensor x, Tensor w, Tensor b, Tensor(a!) rmean, Tensor(b!) rvar, float momentum, float eps, bool relu, Tensor sums) -> (Tensor, Tensor, Tensor, Tensor)");
+  m.def("bn_fwd_train_res_presum(Tensor x, Tensor w, Tensor b, Tensor(a!) rmean, Tensor(b!) rvar, float momentum, float eps, Tensor res, Tensor sums) -> (Tensor, Tensor, Tensor, Tensor)");
   m.def("ln_fwd(Tensor x, Tensor w, Tensor b, float eps) -> (Tensor, Tensor, Tensor)");
   m.def("ln_bwd(Tensor dy, Tensor x, Tensor w, Tensor mean, Tensor invstd) -> (Tensor, Tensor, Tensor)");
   m.def("gelu_fwd(Tensor x) -> Tensor");
@@ -306,6 +323,9 @@ TORCH_LIBRARY_IMPL(dlx_hip, CUDA, m) {
   m.impl("conv2d_fwd", dlx::conv2d_fwd_impl);
   m.impl("conv2d_dgrad", dlx::conv2d_dgrad_impl);
   m.impl("conv2d_wrw", dlx::conv2d_wrw_impl);
+  m.impl("conv2d_fwd_stats", dlx::conv2d_fwd_stats_impl);
+  m.impl("bn_fwd_train_presum", dlx::bn_fwd_train_presum_impl);
+  m.impl("bn_fwd_train_res_presum", dlx::bn_fwd_train_res_presum_impl);
   m.impl("ln_fwd", dlx::ln_fwd_impl);
   m.impl("ln_bwd", dlx::ln_bwd_impl);
   m.impl("gelu_fwd", dlx::gelu_fwd_impl);

@@ -171,6 +171,88 @@ def batch_norm_add_relu(x: torch.Tensor, bn: torch.nn.BatchNorm2d,
 
 
 # ---------------------------------------------------------------------------
+# fully fused conv + BatchNorm (+ReLU / +residual) — the ResNet layer unit
+# ---------------------------------------------------------------------------
+
+class _ConvBNAct(torch.autograd.Function):
+    """conv -> BN (+relu / +residual+relu) with the BN statistics reduced
+    in the conv epilogue (conv2d_fwd_stats): the whole-tensor BN reduce
+    pass disappears.  Backward chains bn_bwd(_res)_m then conv dgrad/wrw.
+    """
+
+    @staticmethod
+    def forward(ctx, x, weight, stride: int, pad: int, bnw, bnb,
+                running_mean, running_var, momentum: float, eps: float,
+                relu_: bool, residual):
+        x = _to_nhwc(x)
+        yc, sums = native().conv2d_fwd_stats(x, weight, stride, pad)
+        if residual is not None:
+            y, mean, invstd, mask = native().bn_fwd_train_res_presum(
+                yc, bnw, bnb, running_mean, running_var, momentum, eps,
+                _to_nhwc(residual), sums)
+        else:
+            y, mean, invstd, mask = native().bn_fwd_train_presum(
+                yc, bnw, bnb, running_mean, running_var, momentum, eps,
+                relu_, sums)
+        ctx.save_for_backward(x, weight, yc, bnw, mean, invstd, mask)
+        ctx.meta = (stride, pad, relu_, residual is not None)
+        return y
+
+    @staticmethod
+    def backward(ctx, grad: torch.Tensor):
+        x, weight, yc, bnw, mean, invstd, mask = ctx.saved_tensors
+        s, p, relu_, has_res = ctx.meta
+        grad = _to_nhwc(grad)
+        dres = None
+        if has_res:
+            dyc, dgamma, dbeta, dres = native().bn_bwd_res_m(
+                grad, yc, bnw, mean, invstd, mask)
+        elif relu_:
+            dyc, dgamma, dbeta = native().bn_bwd_m(
+                grad, yc, bnw, mean, invstd, mask)
+        else:
+            # relu=False never reads its y argument
+            dyc, dgamma, dbeta = native().bn_bwd(
+                grad, yc, bnw, mean, invstd, yc, False)
+        dx = dw = None
+        if ctx.needs_input_grad[0]:
+            dx = native().conv2d_dgrad(dyc, weight, x.shape[2], x.shape[3],
+                                       s, p)
+        if ctx.needs_input_grad[1]:
+            if _os.environ.get("DLX_CONV_WRW", "native") == "native":
+                dw = native().conv2d_wrw(dyc, x, weight.shape[2],
+                                         weight.shape[3], s, p)
+            else:
+                dw = torch.ops.aten.convolution_backward(
+                    dyc, x, weight, None, [s, s], [p, p], [1, 1], False,
+                    [0, 0], 1, [False, True, False])[1]
+        return (dx, dw, None, None, dgamma, dbeta, None, None, None, None,
+                None, dres)
+
+
+def conv_bn_act(x: torch.Tensor, weight: torch.Tensor,
+                bn: torch.nn.BatchNorm2d, stride: int = 1, padding: int = 0,
+                relu: bool = True, residual: Optional[torch.Tensor] = None,
+                training: bool = True) -> torch.Tensor:
+    """The ResNet layer unit: bn(conv(x)) (+residual) (+relu), fused.
+
+    Native path computes the BN channel sums inside the conv epilogue
+    (SURVEY.md §3.2 conv/BN rows collapsed into one producer).  Falls
+    back to the composable ops outside the native gate.
+    """
+    if (training and _use_native(x) and _conv_policy() == "native"
+            and x.dtype == torch.bfloat16
+            and weight.shape[1] % 8 == 0 and weight.shape[0] % 8 == 0):
+        return _ConvBNAct.apply(x, weight, stride, padding, bn.weight,
+                                bn.bias, bn.running_mean, bn.running_var,
+                                bn.momentum, bn.eps, relu, residual)
+    out = conv2d(x, weight, stride, padding)
+    if residual is not None:
+        return batch_norm_add_relu(out, bn, residual, training)
+    return batch_norm_act(out, bn, relu=relu, training=training)
+
+
+# ---------------------------------------------------------------------------
 # fused softmax + cross-entropy (the classifier head)
 # ---------------------------------------------------------------------------
 

@@ -225,3 +225,57 @@ def test_bn_add_relu_fused():
                                atol=2e-1)
     torch.testing.assert_close(bn.bias.grad, bnf.bias.grad, rtol=2e-2,
                                atol=2e-1)
+
+
+@pytest.mark.parametrize("mode", ["relu", "plain", "res"])
+def test_conv_bn_act_fused(mode):
+    """conv+BN(+relu/+residual) with epilogue stats vs fp32 composition."""
+    from deeplearning_cfn_amd.ops import functional as F_amd
+    torch.manual_seed(11)
+    N, Ci, H, W, Co = 4, 64, 14, 14, 128
+    x = torch.randn(N, Ci, H, W, device="cuda").bfloat16()
+    x = x.contiguous(memory_format=torch.channels_last).requires_grad_()
+    wt = (torch.randn(Co, Ci, 3, 3, device="cuda") * (Ci * 9) ** -0.5)
+    wt = wt.bfloat16().contiguous(memory_format=torch.channels_last)
+    wt.requires_grad_()
+    bn = torch.nn.BatchNorm2d(Co).cuda()
+    with torch.no_grad():
+        bn.weight.uniform_(0.5, 1.5)
+        bn.bias.uniform_(-0.5, 0.5)
+    res = None
+    if mode == "res":
+        res = torch.randn(N, Co, H, W, device="cuda").bfloat16()
+        res = res.contiguous(memory_format=torch.channels_last)
+        res.requires_grad_()
+    y = F_amd.conv_bn_act(x, wt, bn, 1, 1, relu=(mode != "plain"),
+                          residual=res, training=True)
+
+    xf = x.detach().float().requires_grad_()
+    wf = wt.detach().float().requires_grad_()
+    bnf = torch.nn.BatchNorm2d(Co).cuda()
+    with torch.no_grad():
+        bnf.weight.copy_(bn.weight)
+        bnf.bias.copy_(bn.bias)
+    ref = torch.nn.functional.batch_norm(
+        torch.nn.functional.conv2d(xf, wf, None, 1, 1), bnf.running_mean,
+        bnf.running_var, bnf.weight, bnf.bias, True, 0.1, 1e-5)
+    if mode == "res":
+        rf = res.detach().float().requires_grad_()
+        ref = ref + rf
+    if mode != "plain":
+        ref = torch.relu(ref)
+    torch.testing.assert_close(y.float(), ref, rtol=5e-2, atol=8e-2)
+    torch.testing.assert_close(bn.running_mean, bnf.running_mean,
+                               rtol=2e-2, atol=2e-2)
+
+    g = torch.randn_like(ref)
+    y.backward(g.bfloat16())
+    ref.backward(g)
+    torch.testing.assert_close(x.grad.float(), xf.grad, rtol=6e-2, atol=0.3)
+    torch.testing.assert_close(wt.grad.float(), wf.grad, rtol=6e-2,
+                               atol=3.0)
+    torch.testing.assert_close(bn.weight.grad, bnf.weight.grad, rtol=3e-2,
+                               atol=0.5)
+    if mode == "res":
+        torch.testing.assert_close(res.grad.float(), rf.grad, rtol=6e-2,
+                                   atol=0.1)
