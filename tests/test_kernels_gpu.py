@@ -229,7 +229,11 @@ def test_bn_add_relu_fused():
 
 @pytest.mark.parametrize("mode", ["relu", "plain", "res"])
 def test_conv_bn_act_fused(mode):
-    """conv+BN(+relu/+residual) with epilogue stats vs fp32 composition."""
+    """Fused conv+BN(+relu/+res) vs the unfused NATIVE composition.
+
+    The reference must share the conv kernel's bf16 numerics: an fp32
+    conv reference rounds differently at relu boundaries and flips the
+    mask on borderline elements (a test artifact, not a defect)."""
     from deeplearning_cfn_amd.ops import functional as F_amd
     torch.manual_seed(11)
     N, Ci, H, W, Co = 4, 64, 14, 14, 128
@@ -239,43 +243,46 @@ def test_conv_bn_act_fused(mode):
     wt = wt.bfloat16().contiguous(memory_format=torch.channels_last)
     wt.requires_grad_()
     bn = torch.nn.BatchNorm2d(Co).cuda()
+    bn2 = torch.nn.BatchNorm2d(Co).cuda()
     with torch.no_grad():
         bn.weight.uniform_(0.5, 1.5)
         bn.bias.uniform_(-0.5, 0.5)
-    res = None
+        bn2.weight.copy_(bn.weight)
+        bn2.bias.copy_(bn.bias)
+    res = res2 = None
     if mode == "res":
         res = torch.randn(N, Co, H, W, device="cuda").bfloat16()
         res = res.contiguous(memory_format=torch.channels_last)
         res.requires_grad_()
+        res2 = res.detach().clone().requires_grad_()
     y = F_amd.conv_bn_act(x, wt, bn, 1, 1, relu=(mode != "plain"),
                           residual=res, training=True)
 
-    xf = x.detach().float().requires_grad_()
-    wf = wt.detach().float().requires_grad_()
-    bnf = torch.nn.BatchNorm2d(Co).cuda()
-    with torch.no_grad():
-        bnf.weight.copy_(bn.weight)
-        bnf.bias.copy_(bn.bias)
-    ref = torch.nn.functional.batch_norm(
-        torch.nn.functional.conv2d(xf, wf, None, 1, 1), bnf.running_mean,
-        bnf.running_var, bnf.weight, bnf.bias, True, 0.1, 1e-5)
+    x2 = x.detach().clone().requires_grad_()
+    w2 = wt.detach().clone().requires_grad_()
+    out = F_amd.conv2d(x2, w2, 1, 1)
     if mode == "res":
-        rf = res.detach().float().requires_grad_()
-        ref = ref + rf
-    if mode != "plain":
-        ref = torch.relu(ref)
-    torch.testing.assert_close(y.float(), ref, rtol=5e-2, atol=8e-2)
-    torch.testing.assert_close(bn.running_mean, bnf.running_mean,
-                               rtol=2e-2, atol=2e-2)
+        ref = F_amd.batch_norm_add_relu(out, bn2, res2, training=True)
+    else:
+        ref = F_amd.batch_norm_act(out, bn2, relu=(mode != "plain"),
+                                   training=True)
+    torch.testing.assert_close(y.float(), ref.float(), rtol=1e-2, atol=1e-2)
+    torch.testing.assert_close(bn.running_mean, bn2.running_mean,
+                               rtol=1e-3, atol=1e-4)
+    torch.testing.assert_close(bn.running_var, bn2.running_var,
+                               rtol=1e-3, atol=1e-4)
 
     g = torch.randn_like(ref)
-    y.backward(g.bfloat16())
+    y.backward(g)
     ref.backward(g)
-    torch.testing.assert_close(x.grad.float(), xf.grad, rtol=6e-2, atol=0.3)
-    torch.testing.assert_close(wt.grad.float(), wf.grad, rtol=6e-2,
-                               atol=3.0)
-    torch.testing.assert_close(bn.weight.grad, bnf.weight.grad, rtol=3e-2,
+    torch.testing.assert_close(x.grad.float(), x2.grad.float(), rtol=2e-2,
+                               atol=2e-2)
+    torch.testing.assert_close(wt.grad.float(), w2.grad.float(), rtol=2e-2,
                                atol=0.5)
+    torch.testing.assert_close(bn.weight.grad, bn2.weight.grad, rtol=1e-3,
+                               atol=1e-2)
+    torch.testing.assert_close(bn.bias.grad, bn2.bias.grad, rtol=1e-3,
+                               atol=1e-2)
     if mode == "res":
-        torch.testing.assert_close(res.grad.float(), rf.grad, rtol=6e-2,
-                                   atol=0.1)
+        torch.testing.assert_close(res.grad.float(), res2.grad.float(),
+                                   rtol=2e-2, atol=2e-2)
