@@ -25,7 +25,7 @@ class SyntheticImages:
                  num_classes: int = 1000, seed: int = 0, rank: int = 0,
                  dtype: torch.dtype = torch.float32,
                  device: str = "cpu", channels_last: bool = False,
-                 pinned: bool = False):
+                 pinned: bool = False, pool: int = 0):
         self.batch_size = batch_size
         self.image_size = image_size
         self.num_classes = num_classes
@@ -33,6 +33,12 @@ class SyntheticImages:
         self.device = device
         self.channels_last = channels_last
         self.pinned = pinned
+        # pool > 0: cycle a FIXED set of batches.  Endless fresh random
+        # labels have ln(num_classes) as their optimum, so a decreasing
+        # loss (memorization) needs a finite pool — the synthetic-data
+        # analogue of the reference's "CIFAR example converges" check.
+        self.pool = pool
+        self._pool_batches = []
         self.gen = torch.Generator().manual_seed(seed * 100003 + rank)
 
     def batch(self) -> Tuple[torch.Tensor, torch.Tensor]:
@@ -51,8 +57,16 @@ class SyntheticImages:
         return x, y
 
     def __iter__(self) -> Iterator[Tuple[torch.Tensor, torch.Tensor]]:
-        while True:
-            yield self.batch()
+        if self.pool > 0:
+            i = 0
+            while True:
+                if len(self._pool_batches) < self.pool:
+                    self._pool_batches.append(self.batch())
+                yield self._pool_batches[i % self.pool]
+                i += 1
+        else:
+            while True:
+                yield self.batch()
 
 
 class SyntheticBertBatches:

@@ -65,6 +65,11 @@ def build_argparser() -> argparse.ArgumentParser:
     p.add_argument("--seq-len", type=int, default=128, help="bert only")
     p.add_argument("--seed", type=int, default=42)
     p.add_argument("--no-prefetch", action="store_true")
+    p.add_argument("--data-pool", type=int, default=0,
+                   help="cycle a fixed pool of N synthetic batches "
+                        "(0 = endless fresh batches); a finite pool makes "
+                        "the loss decrease (memorization) for convergence "
+                        "checks")
     p.add_argument("--checkpoint-every", type=int, default=1,
                    help="epochs between checkpoints (0 = off)")
     p.add_argument("--fault-step", type=int, default=None,
@@ -140,7 +145,8 @@ def train(args: argparse.Namespace) -> dict:
                                  num_classes=ncls, seed=args.seed,
                                  rank=ctx.rank, dtype=dtype,
                                  channels_last=True,
-                                 pinned=ctx.device.type == "cuda")
+                                 pinned=ctx.device.type == "cuda",
+                                 pool=args.data_pool)
     prefetcher = None
     if ctx.device.type == "cuda" and not args.no_prefetch:
         prefetcher = HostPinnedPrefetcher(source, ctx.device)
