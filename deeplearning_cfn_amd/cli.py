@@ -26,7 +26,7 @@ import subprocess
 import sys
 from pathlib import Path
 
-from .contract import EnvContract, make_contract
+from .contract import make_contract
 from .launcher import LaunchSpec, launch_ranks
 from .topo import preflight, probe
 

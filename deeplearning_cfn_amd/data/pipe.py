@@ -11,7 +11,7 @@ from __future__ import annotations
 
 import queue
 import threading
-from typing import Callable, Iterable, Iterator, Optional, Sequence
+from typing import Iterable, Iterator
 
 import torch
 

@@ -13,7 +13,7 @@ fp32 momentum for bf16 params).  Torch fallback keeps CPU tests running.
 
 from __future__ import annotations
 
-from typing import Iterable, List, Optional
+from typing import List, Optional
 
 import torch
 

@@ -17,7 +17,7 @@ import os
 import shutil
 import subprocess
 from dataclasses import dataclass, field
-from typing import List, Optional
+from typing import List
 
 # Tuning constants for the MI355X node.
 XGMI_LINKS_PER_GPU = 7

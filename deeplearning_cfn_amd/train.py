@@ -16,7 +16,7 @@ from __future__ import annotations
 import argparse
 import os
 import sys
-import time
+
 from pathlib import Path
 
 import torch
