@@ -3,7 +3,9 @@
 // Replaces the framework LayerNorm/GELU kernels the reference's stack
 // would use for a transformer (SURVEY.md §3.2 "framework elementwise",
 // BASELINE config #4).  x: [R, D] bf16/fp32, weight/bias fp32, fp32
-// statistics; one 256-thread block per row, vectorized loads.
+// statistics; one 256-thread block per row.  NOTE: loads are scalar
+// (2 B/lane) — 16-B vectorization is a known BERT-path lead
+// (docs/NEXT.md); ResNet does not touch these kernels.
 #include <torch/extension.h>
 #include <ATen/cuda/CUDAContext.h>
 
