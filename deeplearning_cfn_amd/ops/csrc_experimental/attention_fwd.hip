@@ -38,6 +38,7 @@
 
 namespace dlx_experimental {
 
+using dlx::kWave;
 using bf16x8 = __attribute__((ext_vector_type(8))) short;
 using f32x4 = __attribute__((ext_vector_type(4))) float;
 
