@@ -74,8 +74,15 @@ def load_checkpoint(
     state = torch.load(params_path(prefix, epoch), map_location=map_location,
                        weights_only=False)
     model.load_state_dict(state["model"])
-    if optimizer is not None and state.get("optimizer") is not None:
-        optimizer.load_state_dict(state["optimizer"])
+    if optimizer is not None:
+        # Flat optimizers capture fp32 masters at construction; loading the
+        # model copies new values into the flat param buffers, so masters
+        # must be re-derived first.  Checkpointed masters (if saved) then
+        # overwrite these with the exact fp32 values.
+        if hasattr(optimizer, "refresh_masters"):
+            optimizer.refresh_masters()
+        if state.get("optimizer") is not None:
+            optimizer.load_state_dict(state["optimizer"])
     return state
 
 

@@ -60,6 +60,8 @@ def cmd_run(args, cmd: list[str]) -> int:
         cmd=cmd, num_ranks=args.num_ranks, efs_mount=args.efs_mount,
         hostfile=args.hostfile, elastic=args.elastic,
         max_restarts=args.max_restarts,
+        gpus_per_host=args.gpus_per_host,
+        cuda_visible=not args.no_pin_gpus,
         extra_env={"DLX_IMPL": args.impl} if args.impl else {},
     )
     _STATE.mkdir(parents=True, exist_ok=True)
@@ -121,6 +123,12 @@ def main(argv=None) -> int:
     run.add_argument("--elastic", action="store_true")
     run.add_argument("--max-restarts", type=int, default=2)
     run.add_argument("--impl", default=None, help="auto|native|torch")
+    run.add_argument("--gpus-per-host", type=int, default=8,
+                     help="devices per host; rank r is pinned to GPU "
+                          "r %% gpus-per-host via HIP_VISIBLE_DEVICES "
+                          "(1 => all ranks share device 0)")
+    run.add_argument("--no-pin-gpus", action="store_true",
+                     help="do not set per-rank HIP_VISIBLE_DEVICES")
 
     sub.add_parser("kill", help="terminate the last launched run")
 

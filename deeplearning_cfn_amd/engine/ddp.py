@@ -245,6 +245,41 @@ class DataParallel:
                 b.flat.mul_(inv)
                 b.work = None
 
+    # -- observability -----------------------------------------------------
+
+    def comm_bandwidth_probe(self, iters: int = 5) -> List[dict]:
+        """Measure per-bucket all-reduce bus bandwidth vs the xGMI ceiling.
+
+        The metrics contract promises per-bucket
+        all-reduce GB/s against the 153 GB/s/link xGMI ceiling.  In-step
+        collectives overlap backward, so their wall time is not a
+        bandwidth; this probe times ``iters`` blocking all-reduces per
+        bucket (device-synchronised on GPU) and reports the standard
+        ring bus bandwidth ``2*(N-1)/N * bytes / t``.  Collective on all
+        ranks — call from every rank at the same point.
+        """
+        import time
+        out: List[dict] = []
+        if self.world_size == 1 or self.mode == "local":
+            return out
+        sync = (lambda: torch.cuda.synchronize()) if \
+            self.plan.buckets and self.plan.buckets[0].flat.is_cuda else (lambda: None)
+        factor = 2.0 * (self.world_size - 1) / self.world_size
+        for b in self.plan.buckets:
+            scratch = torch.empty_like(b.flat)
+            dist.all_reduce(scratch, group=self.pg)   # warm the channel
+            sync()
+            t0 = time.perf_counter()
+            for _ in range(iters):
+                dist.all_reduce(scratch, group=self.pg)
+            sync()
+            dt = (time.perf_counter() - t0) / iters
+            nbytes = b.flat.numel() * b.flat.element_size()
+            out.append({"bucket": b.index, "mb": nbytes / (1 << 20),
+                        "ms": dt * 1e3,
+                        "busbw_gbps": factor * nbytes / dt / 1e9})
+        return out
+
     # -- parity utilities --------------------------------------------------
 
     def broadcast_parameters(self, src: int = 0) -> None:

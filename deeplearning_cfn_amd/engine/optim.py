@@ -47,6 +47,18 @@ class _FlatOptimBase:
     def zero_grad(self) -> None:
         self.engine.zero_grad()
 
+    def refresh_masters(self) -> None:
+        """Re-derive fp32 master weights from the (bf16) params.
+
+        Needed after ``model.load_state_dict`` — loading copies into the
+        flat param buffers, leaving masters captured at construction time
+        stale.  Checkpoints that carry optimizer state overwrite these
+        again in ``load_state_dict`` (SURVEY.md §6.4 resume contract).
+        """
+        for fp, master in zip(self.flat_params, self.masters):
+            if master is not None:
+                master.copy_(fp.float())
+
     def state_dict(self) -> dict:
         raise NotImplementedError
 
@@ -154,12 +166,20 @@ class FlatAdam(_FlatOptimBase):
             fp.copy_(p32.to(fp.dtype))
 
     def state_dict(self) -> dict:
-        return {"m": self.m, "v": self.v, "t": self.t, "lr": self.lr}
+        return {"m": self.m, "v": self.v, "masters": self.masters,
+                "t": self.t, "lr": self.lr, "betas": self.betas,
+                "eps": self.eps, "weight_decay": self.weight_decay}
 
     def load_state_dict(self, sd: dict) -> None:
         for dst, src in zip(self.m, sd["m"]):
             dst.copy_(src.to(dst.device))
         for dst, src in zip(self.v, sd["v"]):
             dst.copy_(src.to(dst.device))
+        for dst, src in zip(self.masters, sd.get("masters", [])):
+            if dst is not None and src is not None:
+                dst.copy_(src.to(dst.device))
         self.t = sd["t"]
         self.lr = sd["lr"]
+        self.betas = tuple(sd.get("betas", self.betas))
+        self.eps = sd.get("eps", self.eps)
+        self.weight_decay = sd.get("weight_decay", self.weight_decay)

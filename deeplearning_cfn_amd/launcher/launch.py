@@ -114,6 +114,13 @@ def _rank_env(spec: LaunchSpec, contract: EnvContract, rank: int,
     })
     # dmabuf IPC is the only mode the host driver supports (env note).
     env.setdefault("HSA_ENABLE_IPC_MODE_LEGACY", "0")
+    if spec.cuda_visible:
+        # Pin one GPU per rank BEFORE the child imports torch (setting it
+        # inside the process after context init is ineffective).  The
+        # child then sees exactly one device; engine/dist.py's
+        # ``set_device(local_rank % count)`` resolves to device 0.
+        env.setdefault("HIP_VISIBLE_DEVICES",
+                       str(rank % max(spec.gpus_per_host, 1)))
     return env
 
 

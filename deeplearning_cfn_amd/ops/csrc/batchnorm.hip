@@ -556,6 +556,9 @@ bn_fwd_train_res_impl(
     const at::Tensor& res_in) {
   auto x = x_in.contiguous(at::MemoryFormat::ChannelsLast);
   auto res = res_in.contiguous(at::MemoryFormat::ChannelsLast);
+  TORCH_CHECK(res.sizes() == x.sizes() &&
+              res.scalar_type() == x.scalar_type(),
+              "bn_res: residual must match x in shape and dtype");
   auto s = at::cuda::getCurrentCUDAStream();
   const bool bf = x.scalar_type() == at::kBFloat16;
   auto sh = bn_shape(x, bf ? 8 : 4);
@@ -598,6 +601,9 @@ at::Tensor bn_fwd_eval_res_impl(const at::Tensor& x_in, const at::Tensor& w,
                                 const at::Tensor& res_in) {
   auto x = x_in.contiguous(at::MemoryFormat::ChannelsLast);
   auto res = res_in.contiguous(at::MemoryFormat::ChannelsLast);
+  TORCH_CHECK(res.sizes() == x.sizes() &&
+              res.scalar_type() == x.scalar_type(),
+              "bn_res: residual must match x in shape and dtype");
   auto s = at::cuda::getCurrentCUDAStream();
   const bool bf = x.scalar_type() == at::kBFloat16;
   auto sh = bn_shape(x, bf ? 8 : 4);
@@ -814,11 +820,16 @@ bn_fwd_train_res_presum_impl(
     const at::Tensor& res_in, const at::Tensor& sums) {
   auto x = x_in.contiguous(at::MemoryFormat::ChannelsLast);
   auto res = res_in.contiguous(at::MemoryFormat::ChannelsLast);
+  TORCH_CHECK(res.sizes() == x.sizes() &&
+              res.scalar_type() == x.scalar_type(),
+              "bn res presum: residual must match x in shape and dtype");
   auto s = at::cuda::getCurrentCUDAStream();
   const bool bf = x.scalar_type() == at::kBFloat16;
   auto sh = bn_shape(x, bf ? 8 : 4);
   TORCH_CHECK(sh.vec, "bn presum: C % vector width != 0");
   const int nstripes = (int)(sums.numel() / (2 * sh.C));
+  TORCH_CHECK((long)nstripes * 2 * sh.C == sums.numel(),
+              "bn presum: sums size mismatch");
   auto fopts = x.options().dtype(at::kFloat);
   auto save_mean = at::empty({sh.C}, fopts);
   auto save_invstd = at::empty({sh.C}, fopts);

@@ -13,6 +13,8 @@
 // Requires the contiguous k-axis (Cin fwd / Cout dgrad) % 8 == 0; the
 // 3-channel stem stays on the library path (functional.py gate).
 #include <torch/extension.h>
+#include <mutex>
+#include <unordered_map>
 #include <ATen/cuda/CUDAContext.h>
 
 #include "igemm_core.h"
@@ -21,8 +23,11 @@ namespace dlx {
 
 namespace {
 at::Tensor zero_page_for(const at::Tensor& like) {
-  static at::Tensor zp;
-  if (!zp.defined() || zp.device() != like.device())
+  static std::mutex mu;
+  static std::unordered_map<int, at::Tensor> pages;
+  std::lock_guard<std::mutex> lock(mu);
+  auto& zp = pages[like.device().index()];
+  if (!zp.defined())
     zp = at::zeros({16}, like.options().dtype(at::kBFloat16));
   return zp;
 }

@@ -19,6 +19,8 @@
 // geometry: 8 waves (2x4, measured +11%% over 4 at b1024); also serves
 // the Linear weight-grad (gemm_tn_mred_impl) with plain RowGathers.
 #include <torch/extension.h>
+#include <mutex>
+#include <unordered_map>
 #include <ATen/cuda/CUDAContext.h>
 
 #include "igemm_core.h"
@@ -26,6 +28,19 @@
 namespace dlx {
 
 namespace {
+
+// 16-B zero page the gathers read for out-of-image taps.  Per-device,
+// mutex-guarded (a function-local static thrashes across devices and is
+// not thread-safe under first-touch races).
+at::Tensor zero_page_for_wrw(const at::Tensor& like) {
+  static std::mutex mu;
+  static std::unordered_map<int, at::Tensor> pages;
+  std::lock_guard<std::mutex> lock(mu);
+  auto& zp = pages[like.device().index()];
+  if (!zp.defined())
+    zp = at::zeros({16}, like.options().dtype(at::kBFloat16));
+  return zp;
+}
 
 using bf16v4 = __attribute__((ext_vector_type(4))) __bf16;
 using bf16x4 = __attribute__((ext_vector_type(4))) short;
@@ -390,9 +405,7 @@ at::Tensor conv2d_wrw_impl(const at::Tensor& g_in, const at::Tensor& x_in,
   auto outf = (split > 1)
                   ? at::zeros({Rows, Cols}, x.options().dtype(at::kFloat))
                   : at::empty({Rows, Cols}, x.options().dtype(at::kFloat));
-  static at::Tensor zp;  // cached 16-B zero page (per process)
-  if (!zp.defined() || zp.device() != x.device())
-    zp = at::zeros({16}, x.options().dtype(at::kBFloat16));
+  auto zp = zero_page_for_wrw(x);
   auto s = at::cuda::getCurrentCUDAStream();
 
   // A-operand: g rows (m-major, co contiguous); B-operand: x~ im2col gather
@@ -454,9 +467,7 @@ at::Tensor gemm_tn_mred_impl(const at::Tensor& a_in, const at::Tensor& b_in) {
   auto outf = (split > 1)
                   ? at::zeros({Rows, Cols}, A.options().dtype(at::kFloat))
                   : at::empty({Rows, Cols}, A.options().dtype(at::kFloat));
-  static at::Tensor zp;
-  if (!zp.defined() || zp.device() != A.device())
-    zp = at::zeros({16}, A.options().dtype(at::kBFloat16));
+  auto zp = zero_page_for_wrw(A);
   auto s = at::cuda::getCurrentCUDAStream();
   RowGather ag{reinterpret_cast<const uint16_t*>(A.data_ptr()), M, Rows};
   RowGather bg{reinterpret_cast<const uint16_t*>(B.data_ptr()), M, Cols};

@@ -72,6 +72,15 @@ def build_argparser() -> argparse.ArgumentParser:
                         "checks")
     p.add_argument("--checkpoint-every", type=int, default=1,
                    help="epochs between checkpoints (0 = off)")
+    p.add_argument("--no-auto-resume", action="store_true",
+                   help="do not resume from the newest checkpoint under "
+                        "--model-prefix when --load-epoch is absent "
+                        "(default: resume — the reference's elastic-"
+                        "relaunch semantics, SURVEY.md §6.3/§4.4)")
+    p.add_argument("--eval-steps", type=int, default=4,
+                   help="validation batches per epoch (0 = no eval); "
+                        "logged as the MXNet-style Validation-accuracy "
+                        "line (SURVEY.md §6.5)")
     p.add_argument("--fault-step", type=int, default=None,
                    help="test-only: exit(13) at this step on DLX_FAULT_RANK")
     p.add_argument("--profile", action="store_true",
@@ -85,6 +94,46 @@ def build_argparser() -> argparse.ArgumentParser:
 def _dtype(name: str) -> torch.dtype:
     return {"bf16": torch.bfloat16, "bfloat16": torch.bfloat16,
             "fp32": torch.float32, "float32": torch.float32}[name]
+
+
+@torch.no_grad()
+def _evaluate(engine, eval_batches, args, ctx) -> float:
+    """Per-epoch validation pass — the MXNet example's per-epoch
+    ``Validation-accuracy`` analogue.
+
+    ResNet path: top-1 accuracy; BERT path: masked-token prediction
+    accuracy.  Correct/total counts are all-reduced so the printed number
+    is the global validation accuracy, as in the reference's logs.
+    """
+    model = engine.module
+    was_training = model.training
+    model.eval()
+    correct = total = 0
+    for _ in range(args.eval_steps):
+        batch = next(eval_batches)
+        if args.network == "bert":
+            ids, mlm, _ = (t.to(ctx.device) for t in batch)
+            logits = model(ids)                     # [B,S,V], no labels
+            mask = mlm != -100
+            pred = logits.argmax(dim=-1)
+            correct += (pred[mask] == mlm[mask]).sum().item()
+            total += int(mask.sum().item())
+        else:
+            x, y = batch
+            x, y = x.to(ctx.device), y.to(ctx.device)
+            if ctx.device.type == "cuda":
+                x = x.contiguous(memory_format=torch.channels_last)
+            pred = model(x).argmax(dim=-1)
+            correct += (pred == y).sum().item()
+            total += y.numel()
+    if was_training:
+        model.train()
+    import torch.distributed as dist
+    if ctx.world_size > 1 and dist.is_initialized():
+        t = torch.tensor([correct, total], dtype=torch.float64)
+        dist.all_reduce(t)
+        correct, total = t.tolist()
+    return correct / max(total, 1)
 
 
 def train(args: argparse.Namespace) -> dict:
@@ -124,8 +173,15 @@ def train(args: argparse.Namespace) -> dict:
     logger = MetricsLogger(Path(efs) / "metrics.jsonl", rank=ctx.rank)
 
     start_epoch = 0
-    if args.load_epoch is not None:
-        state = ckpt.load_checkpoint(prefix, args.load_epoch, model,
+    load_epoch = args.load_epoch
+    if (load_epoch is None and not args.no_auto_resume
+            and args.checkpoint_every):
+        # Elastic relaunch resumes from the last shared-mount checkpoint
+        # (reference semantics): every rank sees the
+        # same $EFS_MOUNT, so latest_epoch() agrees across ranks.
+        load_epoch = ckpt.latest_epoch(prefix)
+    if load_epoch is not None:
+        state = ckpt.load_checkpoint(prefix, load_epoch, model, opt,
                                      map_location=str(ctx.device))
         start_epoch = state["epoch"]
         engine.broadcast_parameters()
@@ -153,6 +209,33 @@ def train(args: argparse.Namespace) -> dict:
         batches = iter(prefetcher)
     else:
         batches = iter(source)
+
+    # validation source: with a finite memorization pool, "validation" is
+    # the pool itself (held-out random labels are chance forever — the
+    # synthetic analogue of the CIFAR example's accuracy is memorization
+    # accuracy); endless-data runs get a held-out seed.
+    eval_batches = None
+    if args.eval_steps:
+        if args.data_pool > 0 and not is_bert:
+            eval_batches = iter(source)
+        elif is_bert:
+            eval_batches = iter(SyntheticBertBatches(
+                args.batch_size, seq_len=args.seq_len,
+                seed=args.seed + 7919, rank=ctx.rank))
+        else:
+            eval_batches = iter(SyntheticImages(
+                args.batch_size, image_size=image_size, num_classes=ncls,
+                seed=args.seed + 7919, rank=ctx.rank, dtype=dtype,
+                channels_last=True))
+
+    if ctx.world_size > 1 and args.kv_store != "local":
+        # per-bucket all-reduce bandwidth vs the 153 GB/s xGMI link
+        # ceiling (SURVEY.md §6.5 metrics contract)
+        from .topo import XGMI_GBPS_PER_LINK
+        for row in engine.comm_bandwidth_probe():
+            logger.log("comm_bw", bucket=row["bucket"], mb=row["mb"],
+                       ms=row["ms"], busbw_gbps=row["busbw_gbps"],
+                       link_ceiling_gbps=XGMI_GBPS_PER_LINK)
 
     from .ops import functional as F_amd
     fault_rank = int(os.environ.get("DLX_FAULT_RANK", "-1"))
@@ -216,9 +299,17 @@ def train(args: argparse.Namespace) -> dict:
                 timer.start()
                 interval_samples = 0
         timer.stop()
+        if args.eval_steps:
+            acc = _evaluate(engine, eval_batches, args, ctx)
+            if ctx.rank == 0:
+                # the MXNet example's per-epoch line, verbatim shape
+                # (SURVEY.md §6.5): "Epoch[N] Validation-accuracy=V"
+                print(f"Epoch[{epoch}] Validation-accuracy={acc:.6f}",
+                      flush=True)
+            logger.log("eval", epoch=epoch, validation_accuracy=acc)
         if (ctx.is_master and args.checkpoint_every
                 and (epoch + 1) % args.checkpoint_every == 0):
-            ckpt.save_checkpoint(prefix, epoch + 1, model)
+            ckpt.save_checkpoint(prefix, epoch + 1, model, opt)
             logger.log("checkpoint", epoch=epoch + 1, prefix=prefix)
         ctx.barrier()
 
