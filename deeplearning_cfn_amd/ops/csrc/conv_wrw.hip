@@ -102,14 +102,23 @@ DLX_DEV int frag_off(int kk, int cb, int i, int lane) {
          (lane & 15) * 4;
 }
 
+// 128x128 2-phase reduce-major kernel.  Round-2 rework:
+//  * per-slot col state HOISTED out of the m-loop via the gathers'
+//    prep()/at() API (the k-decomposition magic divides run once, not
+//    per tile; the A operand's per-tile address is one 64-bit add);
+//  * split-m writes PARTIAL PLANES (out + slice*Rows*Cols, plain
+//    stores) summed by the host — deterministic (no fp32 atomics, the
+//    ADVICE r1 determinism item) and free of atomic contention on
+//    small outputs (64x64 dW had 768 blocks hammering 16 KB).
 template <class AG, class BG, int WM = 2, int WN = 2>
 __global__ __launch_bounds__(WM* WN * 64) void conv_wrw_kernel(
-    AG ag, BG bg, float* __restrict__ out,  // [BM-rows=co, BN-cols=kf]
+    AG ag, BG bg, float* __restrict__ out,  // [split][Rows][Cols]
     const uint16_t* __restrict__ zero_page, long Mred, long Rows, long Cols,
     int grid_n, int split, long m_per_slice) {
   constexpr int THREADS = WM * WN * 64;
   constexpr int MR = (128 / WM) / 16;
   constexpr int NR = (128 / WN) / 16;
+  constexpr int SPT = kSlotsW / THREADS;  // staging slots per thread
   __shared__ uint16_t lds[2 * 2 * kSlotsW * 8];  // [buf][A|B][slots][8]
 
   int nwg = gridDim.x;
@@ -135,20 +144,30 @@ __global__ __launch_bounds__(WM* WN * 64) void conv_wrw_kernel(
   f32x4 acc[MR][NR] = {};
   const long ntile = (mend - mbeg + kBKw - 1) / kBKw;
 
+  // hoist per-slot loop invariants (slot id is fixed per thread)
+  int mloc_[SPT];
+  typename AG::Pre apre[SPT];
+  typename BG::Pre bpre[SPT];
+#pragma unroll
+  for (int u = 0; u < SPT; ++u) {
+    const int f = u * THREADS + threadIdx.x;
+    int c;
+    slot_to_mc<3>(f, mloc_[u], c);
+    apre[u] = ag.prep(mloc_[u], row0 + c);
+    bpre[u] = bg.prep(mloc_[u], col0 + c);
+  }
+
   auto stage = [&](int buf, long t) {
     const long mt0 = mbeg + t * kBKw;
+    const long mlim = mend - mt0;
     uint16_t* la = lds + (buf * 2 + 0) * kSlotsW * 8;
     uint16_t* lb = lds + (buf * 2 + 1) * kSlotsW * 8;
 #pragma unroll
-    for (int u = 0; u < kSlotsW / THREADS; ++u) {
+    for (int u = 0; u < SPT; ++u) {
       const int f = u * THREADS + threadIdx.x;
-      int mloc, c;
-      slot_to_mc<3>(f, mloc, c);
-      const long m = mt0 + mloc;
-      const uint16_t* sa =
-          (m < mend) ? ag.src(m, row0 + c, zero_page) : zero_page;
-      const uint16_t* sb =
-          (m < mend) ? bg.src(m, col0 + c, zero_page) : zero_page;
+      const bool mv = mloc_[u] < mlim;
+      const uint16_t* sa = ag.at(apre[u], mt0, mloc_[u], mv, zero_page);
+      const uint16_t* sb = bg.at(bpre[u], mt0, mloc_[u], mv, zero_page);
       __builtin_amdgcn_global_load_lds(
           (glb_u32_t*)sa, (lds_u32_t*)(la + (long)f * 8), 16, 0, 0);
       __builtin_amdgcn_global_load_lds(
@@ -197,7 +216,9 @@ __global__ __launch_bounds__(WM* WN * 64) void conv_wrw_kernel(
     buf ^= 1;
   }
 
-  // epilogue: D map col=lane&15, row=(lane>>4)*4+r; atomic accumulate
+  // epilogue: D map col=lane&15, row=(lane>>4)*4+r — plain stores into
+  // this slice's partial plane (host sums planes; deterministic)
+  float* op = out + (long)slice * Rows * Cols;
 #pragma unroll
   for (int mi = 0; mi < MR; ++mi) {
 #pragma unroll
@@ -206,12 +227,8 @@ __global__ __launch_bounds__(WM* WN * 64) void conv_wrw_kernel(
       for (int r = 0; r < 4; ++r) {
         const long row = row0 + wr * (128 / WM) + mi * 16 + l4 * 4 + r;
         const long col = col0 + wc * (128 / WN) + ni * 16 + l15;
-        if (row < Rows && col < Cols) {
-          if (split > 1)
-            unsafeAtomicAdd(&out[row * Cols + col], acc[mi][ni][r]);
-          else
-            out[row * Cols + col] = acc[mi][ni][r];
-        }
+        if (row < Rows && col < Cols)
+          op[row * Cols + col] = acc[mi][ni][r];
       }
     }
   }
@@ -401,41 +418,68 @@ at::Tensor conv2d_wrw_impl(const at::Tensor& g_in, const at::Tensor& x_in,
   long m_per_slice = (p.M + split - 1) / split;
   // round slices to BK so every tile but the last is full
   m_per_slice = ((m_per_slice + kBKw - 1) / kBKw) * kBKw;
+  // rounding can empty trailing slices; every launched slice must write
+  // its partial plane, so shrink split to the covered count
+  split = (p.M + m_per_slice - 1) / m_per_slice;
 
-  auto outf = (split > 1)
-                  ? at::zeros({Rows, Cols}, x.options().dtype(at::kFloat))
-                  : at::empty({Rows, Cols}, x.options().dtype(at::kFloat));
   auto zp = zero_page_for_wrw(x);
   auto s = at::cuda::getCurrentCUDAStream();
 
   // A-operand: g rows (m-major, co contiguous); B-operand: x~ im2col gather
   RowGather ag{reinterpret_cast<const uint16_t*>(g.data_ptr()), p.M,
                (long)p.Co};
-  ConvFwdGather bg{reinterpret_cast<const uint16_t*>(x.data_ptr()), p};
   const int nblocks = gm * gn * (int)split;
   // 8 waves (2x4) measured +11%% over 4 waves at b1024 (profiles/)
   static const bool w8 = []() {
     const char* e = getenv("DLX_WRW_WAVES");
     return !(e && e[0] == '4');
   }();
-  if (big)
+
+  at::Tensor outf;
+  if (big) {
+    // 8-phase experiment path keeps the atomic-accumulate epilogue
+    outf = (split > 1)
+               ? at::zeros({Rows, Cols}, x.options().dtype(at::kFloat))
+               : at::empty({Rows, Cols}, x.options().dtype(at::kFloat));
+    ConvFwdGather bg{reinterpret_cast<const uint16_t*>(x.data_ptr()), p};
     conv_wrw_kernel_8ph<RowGather, ConvFwdGather>
         <<<nblocks, 512, 0, s>>>(
             ag, bg, outf.data_ptr<float>(),
             reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, Rows,
             Cols, gn, (int)split, m_per_slice);
-  else if (w8)
-    conv_wrw_kernel<RowGather, ConvFwdGather, 2, 4>
-        <<<nblocks, 512, 0, s>>>(
+  } else {
+    // deterministic split-m: one partial plane per slice, summed below
+    outf = at::empty({split, Rows, Cols}, x.options().dtype(at::kFloat));
+    const bool plain = p.Kh == 1 && p.Kw == 1 && p.stride == 1 && p.pad == 0;
+    if (plain) {
+      // 1x1/s1: x~ IS x ([M, Ci] row-major) — no im2col address math
+      RowGather bg{reinterpret_cast<const uint16_t*>(x.data_ptr()), p.M,
+                   (long)p.Ci};
+      if (w8)
+        conv_wrw_kernel<RowGather, RowGather, 2, 4><<<nblocks, 512, 0, s>>>(
             ag, bg, outf.data_ptr<float>(),
             reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, Rows,
             Cols, gn, (int)split, m_per_slice);
-  else
-    conv_wrw_kernel<RowGather, ConvFwdGather>
-        <<<nblocks, kThreadsW, 0, s>>>(
+      else
+        conv_wrw_kernel<RowGather, RowGather><<<nblocks, kThreadsW, 0, s>>>(
             ag, bg, outf.data_ptr<float>(),
             reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, Rows,
             Cols, gn, (int)split, m_per_slice);
+    } else {
+      WrwXGather bg{reinterpret_cast<const uint16_t*>(x.data_ptr()), p};
+      if (w8)
+        conv_wrw_kernel<RowGather, WrwXGather, 2, 4><<<nblocks, 512, 0, s>>>(
+            ag, bg, outf.data_ptr<float>(),
+            reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, Rows,
+            Cols, gn, (int)split, m_per_slice);
+      else
+        conv_wrw_kernel<RowGather, WrwXGather><<<nblocks, kThreadsW, 0, s>>>(
+            ag, bg, outf.data_ptr<float>(),
+            reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, Rows,
+            Cols, gn, (int)split, m_per_slice);
+    }
+    if (split > 1) outf = outf.sum(0);
+  }
 
   // [Co, Kf] fp32 -> bf16 tensor with channels_last strides of the weight
   return outf.to(at::kBFloat16)
@@ -463,10 +507,10 @@ at::Tensor gemm_tn_mred_impl(const at::Tensor& a_in, const at::Tensor& b_in) {
   if (split < 1) split = 1;
   long m_per_slice = (M + split - 1) / split;
   m_per_slice = ((m_per_slice + kBKw - 1) / kBKw) * kBKw;
+  split = (M + m_per_slice - 1) / m_per_slice;
 
-  auto outf = (split > 1)
-                  ? at::zeros({Rows, Cols}, A.options().dtype(at::kFloat))
-                  : at::empty({Rows, Cols}, A.options().dtype(at::kFloat));
+  // deterministic split-m partial planes (see conv_wrw_kernel epilogue)
+  auto outf = at::empty({split, Rows, Cols}, A.options().dtype(at::kFloat));
   auto zp = zero_page_for_wrw(A);
   auto s = at::cuda::getCurrentCUDAStream();
   RowGather ag{reinterpret_cast<const uint16_t*>(A.data_ptr()), M, Rows};
@@ -476,7 +520,8 @@ at::Tensor gemm_tn_mred_impl(const at::Tensor& a_in, const at::Tensor& b_in) {
           ag, bg, outf.data_ptr<float>(),
           reinterpret_cast<const uint16_t*>(zp.data_ptr()), M, Rows, Cols,
           gn, (int)split, m_per_slice);
-  return outf.to(at::kBFloat16);
+  if (split > 1) outf = outf.sum(0);
+  return outf.to(at::kBFloat16).view({Rows, Cols});
 }
 
 }  // namespace dlx

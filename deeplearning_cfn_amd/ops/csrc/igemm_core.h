@@ -41,6 +41,19 @@ struct RowGather {
                               const uint16_t* zp) const {
     return (row < rows && k < K) ? base + row * K + k : zp;
   }
+  // Two-phase API for the reduce-major (wrw) kernel: the column index of
+  // a staging slot is loop-invariant across m-tiles, so everything
+  // col-dependent is hoisted into prep(); at() is one 64-bit add.
+  struct Pre { const uint16_t* pbase; };  // nullptr = col out of range
+  DLX_DEV Pre prep(int mloc, long col) const {
+    Pre p;
+    p.pbase = (col < K) ? base + (long)mloc * K + col : nullptr;
+    return p;
+  }
+  DLX_DEV const uint16_t* at(const Pre& p, long mt0, int /*mloc*/,
+                             bool mvalid, const uint16_t* zp) const {
+    return (p.pbase && mvalid) ? p.pbase + mt0 * K : zp;
+  }
 };
 
 // Division by a runtime-constant divisor via magic multiply (round-up
@@ -96,6 +109,46 @@ struct ConvFwdGather {
     const int wi = wo * p.stride - p.pad + dw;
     if (hi < 0 || hi >= p.H || wi < 0 || wi >= p.W) return zp;
     return base + (((long)n * p.H + hi) * p.W + wi) * p.Ci + ci;
+  }
+};
+
+// conv wrw B-operand: x~ im2col with the k-decomposition HOISTED.
+// prep() resolves col=(dh,dw,ci) once per staging slot (loop-invariant
+// across m-tiles); at() only decomposes m=(n,ho,wo) — 2 magic divides
+// instead of 4 plus shorter address math in the hot staging loop.
+struct WrwXGather {
+  const uint16_t* base;
+  ConvShape p;
+  struct Pre {
+    int ci;
+    int dh;   // dh - pad (ready to add to ho*stride)
+    int dw;   // dw - pad
+    int ok;
+  };
+  DLX_DEV Pre prep(int /*mloc*/, long col) const {
+    Pre r;
+    r.ok = col < p.K;
+    const uint32_t k = (uint32_t)(r.ok ? col : 0);
+    const uint32_t dhw = p.dv_ci.div(k);
+    r.ci = (int)k - (int)dhw * p.Ci;
+    const uint32_t dh = p.dv_kw.div(dhw);
+    r.dh = (int)dh - p.pad;
+    r.dw = (int)(dhw - dh * p.Kw) - p.pad;
+    return r;
+  }
+  DLX_DEV const uint16_t* at(const Pre& r, long mt0, int mloc, bool mvalid,
+                             const uint16_t* zp) const {
+    if (!r.ok || !mvalid) return zp;
+    const uint32_t m = (uint32_t)(mt0 + mloc);
+    const uint32_t n = p.dv_hw.div(m);
+    const uint32_t rem = m - n * (uint32_t)(p.Ho * p.Wo);
+    const uint32_t ho = p.dv_wo.div(rem);
+    const uint32_t wo = rem - ho * (uint32_t)p.Wo;
+    const int hi = (int)ho * p.stride + r.dh;
+    const int wi = (int)wo * p.stride + r.dw;
+    if ((unsigned)hi >= (unsigned)p.H || (unsigned)wi >= (unsigned)p.W)
+      return zp;
+    return base + (((long)n * p.H + hi) * p.W + wi) * p.Ci + r.ci;
   }
 };
 

@@ -119,5 +119,31 @@ def test_nccl_ws1_init_and_collectives():
         dist.broadcast(t, src=0)
         torch.cuda.synchronize()
         assert t.sum().item() == 1024
+
+        # hipGraph capture CONTAINING an RCCL collective (the ws>1
+        # variant skips while two-ranks-one-device is refused; this at
+        # least proves RCCL ops capture+replay on this stack)
+        x = torch.ones(1 << 16, device="cuda")
+        buf = torch.empty_like(x)
+
+        def step():
+            buf.copy_(x * 2.0)
+            dist.all_reduce(buf)
+            buf.mul_(0.25)
+
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s):
+            for _ in range(2):
+                step()
+        torch.cuda.current_stream().wait_stream(s)
+        torch.cuda.synchronize()
+        g = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(g):
+            step()
+        for _ in range(3):
+            g.replay()
+        torch.cuda.synchronize()
+        assert torch.allclose(buf, torch.full_like(buf, 0.5)), buf[:4]
     finally:
         dist.destroy_process_group()
