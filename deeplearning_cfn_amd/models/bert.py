@@ -97,7 +97,11 @@ class BertForPretraining(nn.Module):
     def __init__(self, cfg: BertConfig):
         super().__init__()
         self.cfg = cfg
-        self.tok = nn.Embedding(cfg.vocab_size, cfg.hidden)
+        # vocab padded to a multiple of 64 so the weight-tied MLM-head
+        # GEMM meets the MFMA kernel's alignment gate;
+        # pad rows are never labeled, CE simply learns them away
+        self.vocab_padded = (cfg.vocab_size + 63) // 64 * 64
+        self.tok = nn.Embedding(self.vocab_padded, cfg.hidden)
         self.pos = nn.Embedding(cfg.max_seq, cfg.hidden)
         self.typ = nn.Embedding(cfg.type_vocab, cfg.hidden)
         self.ln = FusedLayerNorm(cfg.hidden, eps=cfg.layer_norm_eps)
@@ -129,12 +133,16 @@ class BertForPretraining(nn.Module):
                 nsp_labels: torch.Tensor | None = None):
         x = self.encode(input_ids)
         h = self.mlm_ln(F.gelu(self.mlm_dense(x)))
-        # MLM logits via weight tying with the token embedding
-        logits = F.linear(h, self.tok.weight)
+        # MLM logits via weight tying with the token embedding; routed
+        # through the MFMA GEMM (padded vocab keeps N % 16 == 0)
+        from ..ops import functional as F_amd
+        B, S, H = h.shape
+        logits = F_amd.linear(h.reshape(-1, H), self.tok.weight)
+        logits = logits.reshape(B, S, self.vocab_padded)
         if mlm_labels is None:
             return logits
         loss = F.cross_entropy(
-            logits.float().view(-1, self.cfg.vocab_size),
+            logits.float().view(-1, self.vocab_padded),
             mlm_labels.view(-1), ignore_index=-100)
         if nsp_labels is not None:
             cls = self.nsp(x[:, 0])
