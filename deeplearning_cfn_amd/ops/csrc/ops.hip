@@ -37,6 +37,12 @@ at::Tensor softmax_ce_bwd_impl(const at::Tensor& probs,
 at::Tensor gemm_nt_impl(const at::Tensor& A, const at::Tensor& B,
                         bool out_bf16);
 
+// attention.hip
+std::tuple<at::Tensor, at::Tensor> attn_fwd_impl(const at::Tensor& q,
+                                                 const at::Tensor& k,
+                                                 const at::Tensor& v,
+                                                 double scale);
+
 // batchnorm.hip
 std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
 bn_fwd_train_impl(
@@ -290,6 +296,7 @@ TORCH_LIBRARY(dlx_hip, m) {
   m.def("ln_bwd(Tensor dy, Tensor x, Tensor w, Tensor mean, Tensor invstd) -> (Tensor, Tensor, Tensor)");
   m.def("gelu_fwd(Tensor x) -> Tensor");
   m.def("gelu_bwd(Tensor g, Tensor x) -> Tensor");
+  m.def("attn_fwd(Tensor q, Tensor k, Tensor v, float scale) -> (Tensor, Tensor)");
 }
 
 TORCH_LIBRARY_IMPL(dlx_hip, CUDA, m) {
@@ -330,4 +337,5 @@ TORCH_LIBRARY_IMPL(dlx_hip, CUDA, m) {
   m.impl("ln_bwd", dlx::ln_bwd_impl);
   m.impl("gelu_fwd", dlx::gelu_fwd_impl);
   m.impl("gelu_bwd", dlx::gelu_bwd_impl);
+  m.impl("attn_fwd", dlx::attn_fwd_impl);
 }

@@ -460,3 +460,57 @@ def linear(x: torch.Tensor, weight: torch.Tensor,
             and x.shape[1] % 16 == 0 and weight.shape[0] % 16 == 0):
         return _LinearMFMA.apply(x, weight, bias)
     return F.linear(x, weight, bias)
+
+
+# ---------------------------------------------------------------------------
+# fused attention (flash-style forward kernel; attention.hip)
+# ---------------------------------------------------------------------------
+
+class _AttentionMFMA(torch.autograd.Function):
+    """Fused forward (scores never materialised); the backward recomputes
+    the probabilities from q,k (standard flash recompute composition) and
+    applies the exact attention gradient formulas with batched GEMMs.
+    Memory: O(S^2) only inside backward, transiently.
+    """
+
+    @staticmethod
+    def forward(ctx, q, k, v, scale):
+        o, lse = native().attn_fwd(q, k, v, scale)
+        ctx.save_for_backward(q, k, v, o, lse)
+        ctx.scale = scale
+        return o
+
+    @staticmethod
+    def backward(ctx, grad):
+        q, k, v, o, lse = ctx.saved_tensors
+        scale = ctx.scale
+        grad = grad.contiguous()
+        qf, kf, vf = q.float(), k.float(), v.float()
+        gf = grad.float()
+        # recompute P row-exactly via the saved log-sum-exp
+        s = (qf @ kf.transpose(-1, -2)) * scale          # [B,H,S,S]
+        p = torch.exp(s - lse.unsqueeze(-1))
+        dv = p.transpose(-1, -2) @ gf
+        dp = gf @ vf.transpose(-1, -2)
+        # dS = P * (dP - rowsum(dO * O))  (flash-bwd 'delta' trick)
+        delta = (gf * o.float()).sum(-1, keepdim=True)
+        ds = p * (dp - delta)
+        dq = (ds @ kf) * scale
+        dk = (ds.transpose(-1, -2) @ qf) * scale
+        return dq.to(q.dtype), dk.to(k.dtype), dv.to(v.dtype), None
+
+
+def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
+              scale: Optional[float] = None) -> torch.Tensor:
+    """q,k,v: [B,H,S,Dh].  Fused CDNA4 kernel when Dh==64, S%64==0, bf16
+    on GPU (DLX_ATTN=0 forces the bmm+softmax composition)."""
+    import os
+    if scale is None:
+        scale = q.shape[-1] ** -0.5
+    if (_use_native(q) and q.dtype == torch.bfloat16 and q.shape[-1] == 64
+            and q.shape[-2] % 64 == 0 and os.environ.get("DLX_ATTN") != "0"):
+        return _AttentionMFMA.apply(q, k, v, scale)
+    # composition fallback: explicit batched GEMM + fp32 softmax
+    s = (q @ k.transpose(-1, -2)) * scale
+    p = torch.softmax(s.float(), dim=-1).to(q.dtype)
+    return p @ v
