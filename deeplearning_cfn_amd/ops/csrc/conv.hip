@@ -31,7 +31,75 @@ at::Tensor zero_page_for(const at::Tensor& like) {
     zp = at::zeros({16}, like.options().dtype(at::kBFloat16));
   return zp;
 }
+// ---- small-Cin ("stem") support: channel+width padded copy ------------
+// x [N,Ci<=4,H,W] CL -> x4p [N,H,Wp,4] bf16 (zero pad channels to 4 and
+// width by `pad` left + up to Kwp-1-pad right) so every 16-B im2col slot
+// is memory-contiguous (StemGather, igemm_core.h).
+
+__global__ void stem_pack_kernel(const uint16_t* __restrict__ x,
+                                 uint16_t* __restrict__ x4p, long chunks,
+                                 int H, int W, int Ci, int Wp, int pad) {
+  const long stride_g = (long)gridDim.x * blockDim.x;
+  for (long i = (long)blockIdx.x * blockDim.x + threadIdx.x; i < chunks;
+       i += stride_g) {
+    // i = (n*H + h)*Wp + wp
+    const int wp = (int)(i % Wp);
+    const long nh = i / Wp;
+    const int wi = wp - pad;
+    uint16_t v[4] = {0, 0, 0, 0};
+    if (wi >= 0 && wi < W) {
+      const uint16_t* src = x + (nh * W + wi) * Ci;
+#pragma unroll
+      for (int c = 0; c < 4; ++c)
+        if (c < Ci) v[c] = src[c];
+    }
+    *reinterpret_cast<uint2*>(x4p + i * 4) =
+        *reinterpret_cast<const uint2*>(v);
+  }
+}
+
+// w [Co,Ci,Kh,Kw] -> flat [Co, Kh*Kwp*4] bf16 with dw/channel zero pad
+at::Tensor stem_pack_w(const at::Tensor& w_in, int Kwp) {
+  const long Co = w_in.size(0), Ci = w_in.size(1), Kh = w_in.size(2),
+             Kw = w_in.size(3);
+  auto wx = w_in.permute({0, 2, 3, 1}).contiguous();  // [Co,Kh,Kw,Ci]
+  auto w4 = at::zeros({Co, Kh, (long)Kwp, 4},
+                      w_in.options().dtype(at::kBFloat16));
+  w4.narrow(2, 0, Kw).narrow(3, 0, Ci).copy_(wx);
+  return w4.view({Co, Kh * (long)Kwp * 4});
+}
+
+StemGather make_stem_gather(const uint16_t* x4p, const ConvShape& p,
+                            int Wp, int Kwp) {
+  StemGather g;
+  g.base = x4p;
+  g.M = p.M;
+  g.K = (long)p.Kh * Kwp * 4;
+  g.Ho = p.Ho; g.Wo = p.Wo; g.H = p.H; g.Wp = Wp;
+  g.stride = p.stride; g.pad = p.pad;
+  g.row4 = Kwp * 4;
+  g.dv_wo.init(p.Wo);
+  g.dv_hw.init(p.Ho * p.Wo);
+  g.dv_row.init(g.row4);
+  return g;
+}
+
 }  // namespace
+
+// exported: conv_wrw.hip packs the same x4p for the stem weight grad
+at::Tensor stem_pack_x(const at::Tensor& x_cl, int Wp, int pad) {
+  const int Ci = x_cl.size(1), H = x_cl.size(2), W = x_cl.size(3);
+  const int N = x_cl.size(0);
+  auto x4p = at::empty({(long)N * H * Wp * 4},
+                       x_cl.options().dtype(at::kBFloat16));
+  const long chunks = (long)N * H * Wp;
+  auto s = at::cuda::getCurrentCUDAStream();
+  stem_pack_kernel<<<grid_for(chunks), kBlock, 0, s>>>(
+      reinterpret_cast<const uint16_t*>(x_cl.data_ptr()),
+      reinterpret_cast<uint16_t*>(x4p.data_ptr()), chunks, H, W, Ci, Wp,
+      pad);
+  return x4p;
+}
 
 // x: [N,Ci,H,W] channels_last bf16; w: [Co,Ci,Kh,Kw] bf16 -> [N,Co,Ho,Wo] CL
 at::Tensor conv2d_fwd_impl(const at::Tensor& x_in, const at::Tensor& w_in,
@@ -48,12 +116,29 @@ at::Tensor conv2d_fwd_impl(const at::Tensor& x_in, const at::Tensor& w_in,
   p.K = (long)p.Kh * p.Kw * p.Ci;
   p.M = (long)p.N * p.Ho * p.Wo;
   p.finalize();
-  TORCH_CHECK(p.Ci % 8 == 0, "conv fwd: Cin % 8 != 0 (use library path)");
+  TORCH_CHECK(p.Ci % 8 == 0 || p.Ci <= 4,
+              "conv fwd: Cin must be %8==0 or <=4");
 
   auto y = at::empty({p.N, p.Co, p.Ho, p.Wo},
                      x.options().memory_format(at::MemoryFormat::ChannelsLast));
   auto zp = zero_page_for(x);
   auto s = at::cuda::getCurrentCUDAStream();
+  if (p.Ci <= 4) {
+    // small-Cin stem path: packed x4p + padded-k weights
+    const int Kwp = (p.Kw + 1) & ~1;
+    const int Wp = (p.Wo - 1) * p.stride + Kwp;
+    auto x4p = stem_pack_x(x, Wp, p.pad);
+    auto w4 = stem_pack_w(w_in, Kwp);
+    StemGather ag = make_stem_gather(
+        reinterpret_cast<const uint16_t*>(x4p.data_ptr()), p, Wp, Kwp);
+    RowGather bg4{reinterpret_cast<const uint16_t*>(w4.data_ptr()),
+                  (long)p.Co, ag.K};
+    igemm_launch<StemGather, RowGather, true>(
+        ag, bg4, y.data_ptr(),
+        reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, p.Co, ag.K,
+        s);
+    return y;
+  }
   RowGather bg{reinterpret_cast<const uint16_t*>(w.data_ptr()), p.Co, p.K};
   if (p.Kh == 1 && p.Kw == 1 && p.stride == 1 && p.pad == 0) {
     // 1x1/s1 is a plain GEMM: y[m,co] = x[m,:] @ W[co,:]^T — skip the
@@ -88,14 +173,35 @@ std::tuple<at::Tensor, at::Tensor> conv2d_fwd_stats_impl(
   p.K = (long)p.Kh * p.Kw * p.Ci;
   p.M = (long)p.N * p.Ho * p.Wo;
   p.finalize();
-  TORCH_CHECK(p.Ci % 8 == 0, "conv fwd: Cin % 8 != 0 (use library path)");
+  TORCH_CHECK(p.Ci % 8 == 0 || p.Ci <= 4,
+              "conv fwd: Cin must be %8==0 or <=4");
 
   auto y = at::empty({p.N, p.Co, p.Ho, p.Wo},
                      x.options().memory_format(at::MemoryFormat::ChannelsLast));
-  const long gm = igemm_gm(p.M, p.Co, p.K);
+  // geometry (and so the stats slab size) keys on the EFFECTIVE K —
+  // the stem path pads K to Kh*Kwp*4
+  const long Keff = (p.Ci <= 4)
+                        ? (long)p.Kh * ((p.Kw + 1) & ~1) * 4
+                        : p.K;
+  const long gm = igemm_gm(p.M, p.Co, Keff);
   auto sums = at::empty({gm * 2 * p.Co}, x.options().dtype(at::kFloat));
   auto zp = zero_page_for(x);
   auto s = at::cuda::getCurrentCUDAStream();
+  if (p.Ci <= 4) {
+    const int Kwp = (p.Kw + 1) & ~1;
+    const int Wp = (p.Wo - 1) * p.stride + Kwp;
+    auto x4p = stem_pack_x(x, Wp, p.pad);
+    auto w4 = stem_pack_w(w_in, Kwp);
+    StemGather ag = make_stem_gather(
+        reinterpret_cast<const uint16_t*>(x4p.data_ptr()), p, Wp, Kwp);
+    RowGather bg4{reinterpret_cast<const uint16_t*>(w4.data_ptr()),
+                  (long)p.Co, ag.K};
+    igemm_launch_stats<StemGather, RowGather>(
+        ag, bg4, y.data_ptr(),
+        reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, p.Co, ag.K,
+        s, sums.data_ptr<float>());
+    return {y, sums};
+  }
   RowGather bg{reinterpret_cast<const uint16_t*>(w.data_ptr()), p.Co, p.K};
   if (p.Kh == 1 && p.Kw == 1 && p.stride == 1 && p.pad == 0) {
     RowGather ag{reinterpret_cast<const uint16_t*>(x.data_ptr()), p.M, p.K};

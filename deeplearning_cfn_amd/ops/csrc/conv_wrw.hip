@@ -379,6 +379,64 @@ __global__ __launch_bounds__(512) void conv_wrw_kernel_8ph(
 
 }  // namespace
 
+// conv.hip: packs x [N,Ci<=4,H,W] CL into the stem layout [N,H,Wp,4]
+at::Tensor stem_pack_x(const at::Tensor& x_cl, int Wp, int pad);
+
+namespace {
+
+// small-Cin stem weight grad: reduce-major GEMM over the packed x4p
+// (StemWrwGather, igemm_core.h); dW columns are (dh, dwp, c4) and the
+// dwp/channel pads are sliced away on the host.
+at::Tensor conv2d_wrw_stem(const at::Tensor& g, const at::Tensor& x,
+                           ConvShape& p) {
+  const int Kwp = (p.Kw + 1) & ~1;
+  const int Wp = (p.Wo - 1) * p.stride + Kwp;
+  auto x4p = stem_pack_x(x, Wp, p.pad);
+
+  StemWrwGather bg;
+  bg.base = reinterpret_cast<const uint16_t*>(x4p.data_ptr());
+  bg.M = p.M;
+  bg.K = (long)p.Kh * Kwp * 4;
+  bg.Ho = p.Ho; bg.Wo = p.Wo; bg.H = p.H; bg.Wp = Wp;
+  bg.stride = p.stride; bg.pad = p.pad;
+  bg.row4 = Kwp * 4;
+  bg.dv_wo.init(p.Wo);
+  bg.dv_hw.init(p.Ho * p.Wo);
+  bg.dv_row.init(bg.row4);
+
+  const long Rows = p.Co, Cols = bg.K;
+  const int gm = (int)((Rows + kBMw - 1) / kBMw);
+  const int gn = (int)((Cols + kBNw - 1) / kBNw);
+  long split = 768 / ((long)gm * gn);
+  const long max_split = (p.M + kBKw - 1) / kBKw;
+  if (split > max_split) split = max_split;
+  if (split < 1) split = 1;
+  long m_per_slice = (p.M + split - 1) / split;
+  m_per_slice = ((m_per_slice + kBKw - 1) / kBKw) * kBKw;
+  split = (p.M + m_per_slice - 1) / m_per_slice;
+
+  auto outf = at::empty({split, Rows, Cols}, x.options().dtype(at::kFloat));
+  auto zp = zero_page_for_wrw(x);
+  auto s = at::cuda::getCurrentCUDAStream();
+  RowGather ag{reinterpret_cast<const uint16_t*>(g.data_ptr()), p.M,
+               (long)p.Co};
+  conv_wrw_kernel<RowGather, StemWrwGather, 2, 4>
+      <<<gm * gn * (int)split, 512, 0, s>>>(
+          ag, bg, outf.data_ptr<float>(),
+          reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, Rows,
+          Cols, gn, (int)split, m_per_slice);
+  auto summed = (split > 1) ? outf.sum(0) : outf.view({Rows, Cols});
+  // [Co, Kh, Kwp, 4] -> slice off the pads -> dW with CL strides
+  return summed.to(at::kBFloat16)
+      .view({(long)p.Co, (long)p.Kh, (long)Kwp, 4})
+      .narrow(2, 0, p.Kw)
+      .narrow(3, 0, p.Ci)
+      .contiguous()
+      .permute({0, 3, 1, 2});
+}
+
+}  // namespace
+
 // g: [N,Co,Ho,Wo] CL bf16;  x: [N,Ci,H,W] CL bf16 ->
 // dW bf16 with channels_last strides of [Co,Ci,Kh,Kw]
 at::Tensor conv2d_wrw_impl(const at::Tensor& g_in, const at::Tensor& x_in,
@@ -395,8 +453,11 @@ at::Tensor conv2d_wrw_impl(const at::Tensor& g_in, const at::Tensor& x_in,
   p.K = (long)p.Kh * p.Kw * p.Ci;   // fwd-gather k-extent (= kf here)
   p.M = (long)p.N * p.Ho * p.Wo;    // reduction extent
   p.finalize();
-  TORCH_CHECK(p.Ci % 8 == 0, "wrw: Cin % 8 != 0 (use library path)");
+  TORCH_CHECK(p.Ci % 8 == 0 || p.Ci <= 4,
+              "wrw: Cin must be %8==0 or <=4");
   TORCH_CHECK(p.Co % 8 == 0, "wrw: Cout % 8 != 0 (use library path)");
+
+  if (p.Ci <= 4) return conv2d_wrw_stem(g, x, p);
 
   const long Rows = p.Co, Cols = p.K;
   // Measured (profiles/): the 128^2 4-wave tile beats the 256^2 8-phase

@@ -152,6 +152,68 @@ struct WrwXGather {
   }
 };
 
+// Small-Cin ("stem") conv gathers over a channel+width padded copy of x:
+// x4p [N, H, Wp=(Wo-1)*stride+Kwp, 4] bf16, where Kwp = Kw rounded up to
+// even and channels padded to 4 with zeros (pack_stem kernel).  The k
+// axis is (dh, dwp, c4) with a 16-B slot = 2 adjacent dwp taps x 4
+// channels — CONTIGUOUS in x4p (the property plain NHWC c=3 lacks), so
+// glds staging works unchanged.  Width padding bakes the wi bounds in;
+// only the hi bound remains a zero-page redirect.  (VERDICT r1 #4.)
+struct StemGather {
+  const uint16_t* base;  // x4p
+  long M, K;             // K = Kh * Kwp * 4
+  int Ho, Wo, H, Wp, stride, pad, row4;  // row4 = Kwp*4
+  U32Div dv_wo, dv_hw, dv_row;
+  DLX_DEV const uint16_t* src(long m, long k, const uint16_t* zp) const {
+    if (m >= M || k >= K) return zp;
+    const uint32_t n = dv_hw.div((uint32_t)m);
+    const uint32_t rem = (uint32_t)m - n * (uint32_t)(Ho * Wo);
+    const uint32_t ho = dv_wo.div(rem);
+    const uint32_t wo = rem - ho * (uint32_t)Wo;
+    const uint32_t dh = dv_row.div((uint32_t)k);
+    const uint32_t krem = (uint32_t)k - dh * (uint32_t)row4;  // dwp*4 + c
+    const int hi = (int)ho * stride - pad + (int)dh;
+    if ((unsigned)hi >= (unsigned)H) return zp;
+    // col in x4p = wo*stride + dwp (pad folded into the packing)
+    return base + (((long)n * H + hi) * Wp + (long)wo * stride) * 4 + krem;
+  }
+};
+
+// wrw twin with the hoisted two-phase API (reduce-major kernel B operand)
+struct StemWrwGather {
+  const uint16_t* base;  // x4p
+  long M, K;
+  int Ho, Wo, H, Wp, stride, pad, row4;
+  U32Div dv_wo, dv_hw, dv_row;
+  struct Pre {
+    int dh;    // dh - pad (ready to add to ho*stride)
+    int krem;  // dwp*4 + c
+    int ok;
+  };
+  DLX_DEV Pre prep(int /*mloc*/, long col) const {
+    Pre r;
+    r.ok = col < K;
+    const uint32_t k = (uint32_t)(r.ok ? col : 0);
+    const uint32_t dh = dv_row.div(k);
+    r.krem = (int)(k - dh * (uint32_t)row4);
+    r.dh = (int)dh - pad;
+    return r;
+  }
+  DLX_DEV const uint16_t* at(const Pre& r, long mt0, int mloc, bool mvalid,
+                             const uint16_t* zp) const {
+    if (!r.ok || !mvalid) return zp;
+    const uint32_t m = (uint32_t)(mt0 + mloc);
+    const uint32_t n = dv_hw.div(m);
+    const uint32_t rem = m - n * (uint32_t)(Ho * Wo);
+    const uint32_t ho = dv_wo.div(rem);
+    const uint32_t wo = rem - ho * (uint32_t)Wo;
+    const int hi = (int)ho * stride + r.dh;
+    if ((unsigned)hi >= (unsigned)H) return zp;
+    return base + (((long)n * H + hi) * Wp + (long)wo * stride) * 4 +
+           r.krem;
+  }
+};
+
 // conv dgrad: row m=(n,hi,wi) over INPUT pixels, k=(dh,dw,co) from grad
 struct ConvDgradGather {
   const uint16_t* base;

@@ -14,6 +14,8 @@ SHAPES = [
     (4, 128, 28, 28, 128, 3, 2, 1),     # strided 3x3
     (2, 2048, 7, 7, 512, 1, 1, 0),      # deep narrow
     (2, 16, 32, 32, 16, 3, 1, 1),       # CIFAR shape
+    (2, 3, 224, 224, 64, 7, 2, 3),      # ImageNet stem (small-Cin path)
+    (2, 3, 32, 32, 16, 3, 1, 1),        # CIFAR stem
 ]
 
 
