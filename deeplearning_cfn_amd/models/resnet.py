@@ -29,7 +29,8 @@ class DLXConv2d(nn.Conv2d):
     """Conv2d dispatching to the implicit-GEMM MFMA kernel (ops/conv.hip).
 
     Falls back to the library path for shapes outside the native surface
-    (Cin%8!=0, e.g. the 3-channel stem) or when DLX_CONV=torch.
+    (Cin%8!=0 and Cin>4 — the 3-channel stems use the packed small-Cin
+    path) or when DLX_CONV=torch.
     """
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:

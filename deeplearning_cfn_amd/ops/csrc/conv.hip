@@ -304,6 +304,16 @@ at::Tensor conv2d_dgrad_impl(const at::Tensor& g_in, const at::Tensor& w_in,
         reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, p.Ci, p.K, s);
     return dx;
   }
+  if (p.stride == 1) {
+    // no /stride divides or exactness checks in the gather
+    ConvDgradS1Gather ag{reinterpret_cast<const uint16_t*>(g.data_ptr()),
+                         p};
+    igemm_launch<ConvDgradS1Gather, RowGather, true>(
+        ag, bg, dx.data_ptr(),
+        reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, p.Ci, p.K,
+        s);
+    return dx;
+  }
   ConvDgradGather ag{reinterpret_cast<const uint16_t*>(g.data_ptr()), p};
   igemm_launch<ConvDgradGather, RowGather, true>(
       ag, bg, dx.data_ptr(),

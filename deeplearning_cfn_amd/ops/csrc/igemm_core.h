@@ -54,6 +54,18 @@ struct RowGather {
                              bool mvalid, const uint16_t* zp) const {
     return (p.pbase && mvalid) ? p.pbase + mt0 * K : zp;
   }
+  // Output-major twin (igemm_kernel): a staging slot's ROW is fixed
+  // across k-tiles — hoist the row base, oat() is one add + bound.
+  struct OPre { const uint16_t* rowbase; };
+  DLX_DEV OPre oprep(long row) const {
+    OPre p;
+    p.rowbase = (row < rows) ? base + row * K : nullptr;
+    return p;
+  }
+  DLX_DEV const uint16_t* oat(const OPre& p, long k,
+                              const uint16_t* zp) const {
+    return (p.rowbase && k < K) ? p.rowbase + k : zp;
+  }
 };
 
 // Division by a runtime-constant divisor via magic multiply (round-up
@@ -109,6 +121,40 @@ struct ConvFwdGather {
     const int wi = wo * p.stride - p.pad + dw;
     if (hi < 0 || hi >= p.H || wi < 0 || wi >= p.W) return zp;
     return base + (((long)n * p.H + hi) * p.W + wi) * p.Ci + ci;
+  }
+  // output-major hoist: m=(n,ho,wo) is fixed per staging slot
+  struct OPre {
+    long nbase;  // n*H, or -1 when m out of range
+    int hi0;     // ho*stride - pad
+    int wi0;     // wo*stride - pad
+  };
+  DLX_DEV OPre oprep(long m) const {
+    OPre r;
+    if (m >= p.M) {
+      r.nbase = -1;
+      return r;
+    }
+    const uint32_t n = p.dv_hw.div((uint32_t)m);
+    const uint32_t rem = (uint32_t)m - n * (uint32_t)(p.Ho * p.Wo);
+    const uint32_t ho = p.dv_wo.div(rem);
+    const uint32_t wo = rem - ho * (uint32_t)p.Wo;
+    r.nbase = (long)n * p.H;
+    r.hi0 = (int)ho * p.stride - p.pad;
+    r.wi0 = (int)wo * p.stride - p.pad;
+    return r;
+  }
+  DLX_DEV const uint16_t* oat(const OPre& r, long k,
+                              const uint16_t* zp) const {
+    if (r.nbase < 0 || k >= p.K) return zp;
+    const uint32_t dhw = p.dv_ci.div((uint32_t)k);
+    const int ci = (int)k - (int)dhw * p.Ci;
+    const uint32_t dh = p.dv_kw.div(dhw);
+    const int dw = (int)(dhw - dh * p.Kw);
+    const int hi = r.hi0 + (int)dh;
+    const int wi = r.wi0 + dw;
+    if ((unsigned)hi >= (unsigned)p.H || (unsigned)wi >= (unsigned)p.W)
+      return zp;
+    return base + ((r.nbase + hi) * p.W + wi) * p.Ci + ci;
   }
 };
 
@@ -177,6 +223,35 @@ struct StemGather {
     // col in x4p = wo*stride + dwp (pad folded into the packing)
     return base + (((long)n * H + hi) * Wp + (long)wo * stride) * 4 + krem;
   }
+  struct OPre {
+    long nbase;    // n*H, or -1 when m out of range
+    long colbase;  // wo*stride*4
+    int hi0;       // ho*stride - pad
+  };
+  DLX_DEV OPre oprep(long m) const {
+    OPre r;
+    if (m >= M) {
+      r.nbase = -1;
+      return r;
+    }
+    const uint32_t n = dv_hw.div((uint32_t)m);
+    const uint32_t rem = (uint32_t)m - n * (uint32_t)(Ho * Wo);
+    const uint32_t ho = dv_wo.div(rem);
+    const uint32_t wo = rem - ho * (uint32_t)Wo;
+    r.nbase = (long)n * H;
+    r.colbase = (long)wo * stride * 4;
+    r.hi0 = (int)ho * stride - pad;
+    return r;
+  }
+  DLX_DEV const uint16_t* oat(const OPre& r, long k,
+                              const uint16_t* zp) const {
+    if (r.nbase < 0 || k >= K) return zp;
+    const uint32_t dh = dv_row.div((uint32_t)k);
+    const uint32_t krem = (uint32_t)k - dh * (uint32_t)row4;
+    const int hi = r.hi0 + (int)dh;
+    if ((unsigned)hi >= (unsigned)H) return zp;
+    return base + ((r.nbase + hi) * Wp) * 4 + r.colbase + krem;
+  }
 };
 
 // wrw twin with the hoisted two-phase API (reduce-major kernel B operand)
@@ -237,6 +312,97 @@ struct ConvDgradGather {
     if (ho >= p.Ho || wo >= p.Wo) return zp;
     return base + (((long)n * p.Ho + ho) * p.Wo + wo) * p.Co + co;
   }
+  struct OPre {
+    long nbase;  // n*Ho, or -1 when m out of range
+    int ht0;     // hi + pad
+    int wt0;     // wi + pad
+  };
+  DLX_DEV OPre oprep(long m) const {
+    OPre r;
+    if (m >= p.M) {
+      r.nbase = -1;
+      return r;
+    }
+    const uint32_t n = p.dv_hw_in.div((uint32_t)m);
+    const uint32_t rem = (uint32_t)m - n * (uint32_t)(p.H * p.W);
+    const uint32_t hi = p.dv_w.div(rem);
+    const uint32_t wi = rem - hi * (uint32_t)p.W;
+    r.nbase = (long)n * p.Ho;
+    r.ht0 = (int)hi + p.pad;
+    r.wt0 = (int)wi + p.pad;
+    return r;
+  }
+  DLX_DEV const uint16_t* oat(const OPre& r, long k,
+                              const uint16_t* zp) const {
+    if (r.nbase < 0 || k >= p.K) return zp;
+    const uint32_t dhw = p.dv_co.div((uint32_t)k);
+    const int co = (int)k - (int)dhw * p.Co;
+    const uint32_t dh = p.dv_kw.div(dhw);
+    const int dw = (int)(dhw - dh * p.Kw);
+    const int ht = r.ht0 - (int)dh;
+    const int wt = r.wt0 - dw;
+    if (ht < 0 || wt < 0) return zp;
+    const int ho = (int)p.dv_s.div((uint32_t)ht);
+    const int wo = (int)p.dv_s.div((uint32_t)wt);
+    if (ho * p.stride != ht || wo * p.stride != wt) return zp;
+    if (ho >= p.Ho || wo >= p.Wo) return zp;
+    return base + ((r.nbase + ho) * p.Wo + wo) * p.Co + co;
+  }
+};
+
+// stride-1 dgrad: ho == ht, wo == wt — no /stride divides, no exactness
+// checks (NEXT.md r1 dgrad lead)
+struct ConvDgradS1Gather {
+  const uint16_t* base;
+  ConvShape p;
+  DLX_DEV const uint16_t* src(long m, long k, const uint16_t* zp) const {
+    if (m >= p.M || k >= p.K) return zp;
+    const int n = (int)p.dv_hw_in.div((uint32_t)m);
+    const int rem = (int)m - n * (p.H * p.W);
+    const int hi = (int)p.dv_w.div((uint32_t)rem);
+    const int wi = rem - hi * p.W;
+    const int dhw = (int)p.dv_co.div((uint32_t)k);
+    const int co = (int)k - dhw * p.Co;
+    const int dh = (int)p.dv_kw.div((uint32_t)dhw);
+    const int dw = dhw - dh * p.Kw;
+    const int ho = hi + p.pad - dh;
+    const int wo = wi + p.pad - dw;
+    if ((unsigned)ho >= (unsigned)p.Ho || (unsigned)wo >= (unsigned)p.Wo)
+      return zp;
+    return base + (((long)n * p.Ho + ho) * p.Wo + wo) * p.Co + co;
+  }
+  struct OPre {
+    long nbase;
+    int ht0, wt0;  // hi + pad, wi + pad
+  };
+  DLX_DEV OPre oprep(long m) const {
+    OPre r;
+    if (m >= p.M) {
+      r.nbase = -1;
+      return r;
+    }
+    const uint32_t n = p.dv_hw_in.div((uint32_t)m);
+    const uint32_t rem = (uint32_t)m - n * (uint32_t)(p.H * p.W);
+    const uint32_t hi = p.dv_w.div(rem);
+    const uint32_t wi = rem - hi * (uint32_t)p.W;
+    r.nbase = (long)n * p.Ho;
+    r.ht0 = (int)hi + p.pad;
+    r.wt0 = (int)wi + p.pad;
+    return r;
+  }
+  DLX_DEV const uint16_t* oat(const OPre& r, long k,
+                              const uint16_t* zp) const {
+    if (r.nbase < 0 || k >= p.K) return zp;
+    const uint32_t dhw = p.dv_co.div((uint32_t)k);
+    const int co = (int)k - (int)dhw * p.Co;
+    const uint32_t dh = p.dv_kw.div(dhw);
+    const int dw = (int)(dhw - dh * p.Kw);
+    const int ho = r.ht0 - (int)dh;
+    const int wo = r.wt0 - dw;
+    if ((unsigned)ho >= (unsigned)p.Ho || (unsigned)wo >= (unsigned)p.Wo)
+      return zp;
+    return base + ((r.nbase + ho) * p.Wo + wo) * p.Co + co;
+  }
 };
 
 // stride-2 dgrad, parity-decomposed: class (a,b) covers input pixels
@@ -266,6 +432,40 @@ struct ConvDgradS2Gather {
     const int ho = ht >> 1, wo = wt >> 1;
     if (ho >= p.Ho || wo >= p.Wo) return zp;
     return base + (((long)n * p.Ho + ho) * p.Wo + wo) * p.Co + co;
+  }
+  struct OPre {
+    long nbase;  // n*Ho, or -1 when m out of range
+    int hti;     // (2i+a) + pad
+    int wtj;     // (2j+b) + pad
+  };
+  DLX_DEV OPre oprep(long m) const {
+    OPre r;
+    if (m >= M2) {
+      r.nbase = -1;
+      return r;
+    }
+    const uint32_t n = dv_hwb.div((uint32_t)m);
+    const uint32_t rem = (uint32_t)m - n * (uint32_t)HaWb;
+    const uint32_t i = dv_wb.div(rem);
+    const uint32_t j = rem - i * (uint32_t)Wb;
+    r.nbase = (long)n * p.Ho;
+    r.hti = (int)(2 * i) + a + p.pad;
+    r.wtj = (int)(2 * j) + b + p.pad;
+    return r;
+  }
+  DLX_DEV const uint16_t* oat(const OPre& r, long k,
+                              const uint16_t* zp) const {
+    if (r.nbase < 0 || k >= K2) return zp;
+    const uint32_t dij = p.dv_co.div((uint32_t)k);
+    const int co = (int)k - (int)dij * p.Co;
+    const uint32_t di = dv_nkw.div(dij);
+    const int dj = (int)(dij - di * nKw);
+    const int ht = r.hti - (pa + 2 * (int)di);
+    const int wt = r.wtj - (pb + 2 * dj);
+    if (ht < 0 || wt < 0) return zp;
+    const int ho = ht >> 1, wo = wt >> 1;
+    if (ho >= p.Ho || wo >= p.Wo) return zp;
+    return base + ((r.nbase + ho) * p.Wo + wo) * p.Co + co;
   }
 };
 
@@ -377,25 +577,40 @@ __global__ __launch_bounds__(WM* WN * 64) void igemm_kernel(
   f32x4 acc[MR][NR] = {};
   const long ntile = (K + BK - 1) / BK;
 
+  // hoist per-slot row state out of the k loop (rows are fixed per slot;
+  // only k advances) — the ConvFwd/Dgrad m-decomposition divides run
+  // once instead of per tile
+  int kca[GA], kcb[GB];
+  typename AG::OPre apre[GA];
+  typename BG::OPre bpre[GB];
+#pragma unroll
+  for (int i = 0; i < GA; ++i) {
+    const int f = i * THREADS + threadIdx.x;
+    kca[i] = f / BM;
+    apre[i] = ag.oprep(m0 + (f - kca[i] * BM));
+  }
+#pragma unroll
+  for (int i = 0; i < GB; ++i) {
+    const int f = i * THREADS + threadIdx.x;
+    kcb[i] = f / BN;
+    bpre[i] = bg.oprep(n0 + (f - kcb[i] * BN));
+  }
+
   auto stage = [&](int buf, long t) {
     const long k0 = t * BK;
     uint16_t* la = lsA + buf * SLOTS_A * 8;
     uint16_t* lb = lsB + buf * SLOTS_B * 8;
 #pragma unroll
     for (int i = 0; i < GA; ++i) {
-      int f = i * THREADS + threadIdx.x;
-      int kc = f / BM;
-      int m = f - kc * BM;
-      const uint16_t* sa = ag.src(m0 + m, k0 + kc * 8, zero_page);
+      const int f = i * THREADS + threadIdx.x;
+      const uint16_t* sa = ag.oat(apre[i], k0 + kca[i] * 8, zero_page);
       __builtin_amdgcn_global_load_lds(
           (glb_u32_t*)sa, (lds_u32_t*)(la + (long)f * 8), 16, 0, 0);
     }
 #pragma unroll
     for (int i = 0; i < GB; ++i) {
-      int f = i * THREADS + threadIdx.x;
-      int kc = f / BN;
-      int n = f - kc * BN;
-      const uint16_t* sb = bg.src(n0 + n, k0 + kc * 8, zero_page);
+      const int f = i * THREADS + threadIdx.x;
+      const uint16_t* sb = bg.oat(bpre[i], k0 + kcb[i] * 8, zero_page);
       __builtin_amdgcn_global_load_lds(
           (glb_u32_t*)sb, (lds_u32_t*)(lb + (long)f * 8), 16, 0, 0);
     }
@@ -512,6 +727,20 @@ __global__ __launch_bounds__(512) void igemm_kernel_8ph(
   f32x4 acc[MR][NR] = {};
   const long ntile = (K + BK - 1) / BK;
 
+  // hoisted per-slot row state (see igemm_kernel): 4 slots per thread
+  // per operand (units x 2 glds)
+  int kc8[4];
+  typename AG::OPre apre[4];
+  typename BG::OPre bpre[4];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const int f = j * THREADS + threadIdx.x;
+    kc8[j] = f / BM;
+    const int r = f - kc8[j] * BM;
+    apre[j] = ag.oprep(m0 + r);
+    bpre[j] = bg.oprep(n0 + r);  // BN == BM == 256
+  }
+
   // stage unit u of tile t into buffer `buf`:
   //   u=0: A k-half 0 (glds instrs 0,1) ; u=1: B k-half 0
   //   u=2: A k-half 1 (instrs 2,3)      ; u=3: B k-half 1
@@ -522,12 +751,11 @@ __global__ __launch_bounds__(512) void igemm_kernel_8ph(
     uint16_t* l0 = lds + (buf * 2 + (isB ? 1 : 0)) * SLOTS * 8;
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      int f = (half * 2 + i) * THREADS + threadIdx.x;
-      int kc = f / BM;
-      int r = f - kc * BM;
-      const uint16_t* src =
-          isB ? bg.src(n0 + r, k0 + kc * 8, zero_page)
-              : ag.src(m0 + r, k0 + kc * 8, zero_page);
+      const int j = half * 2 + i;
+      const int f = j * THREADS + threadIdx.x;
+      const long k = k0 + kc8[j] * 8;
+      const uint16_t* src = isB ? bg.oat(bpre[j], k, zero_page)
+                                : ag.oat(apre[j], k, zero_page);
       __builtin_amdgcn_global_load_lds(
           (glb_u32_t*)src, (lds_u32_t*)(l0 + (long)f * 8), 16, 0, 0);
     }
