@@ -222,6 +222,242 @@ __global__ __launch_bounds__(256) void attn_fwd_kernel(
   }
 }
 
+// ---------------------------------------------------------------------
+// Fused attention BACKWARD (flash-bwd, recompute P^T in-kernel).
+//
+// One block per (b*h, 64-key tile): K/V rows live in registers per wave
+// (16 keys each), Q/dO tiles stream through LDS in BOTH the k-subtile
+// image (dh-contiguous fragments) and the transpose image
+// (row-contiguous fragments via ds_read_b64_tr_b16).  Computing S^T
+// (= K Q^T) instead of S puts P^T and dP^T in the SAME C-map, so
+// dS^T = P^T*(dP^T - delta)*scale is lane-local.  dK/dV accumulate in
+// registers and store plainly (each block owns its keys); dQ^T partials
+// go to a fp32 buffer with atomics (S/64 contributors per element —
+// NOT bitwise deterministic; DLX_ATTN=0 gives the deterministic
+// composition).  5 MFMA stages per (i,j) tile pair:
+//   S^T = K Q^T;  dV += P^T dO;  dP^T = V dO^T;  dK += dS^T Q;
+//   dQ^T += K^T dS^T  (block-level dS^T transpose image).
+// ---------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void attn_bwd_kernel(
+    const uint16_t* __restrict__ q, const uint16_t* __restrict__ k,
+    const uint16_t* __restrict__ v, const uint16_t* __restrict__ dout,
+    const float* __restrict__ lse, const float* __restrict__ delta,
+    uint16_t* __restrict__ dk_out, uint16_t* __restrict__ dv_out,
+    float* __restrict__ dq_f32, int S, float scale) {
+  constexpr int kSlots = kKT * kDh / 8;  // 512 per image
+  __shared__ uint16_t lq[kSlots * 8];    // Q k-subtile image
+  __shared__ uint16_t lqt[kSlots * 8];   // Q transpose image
+  __shared__ uint16_t lo[kSlots * 8];    // dO k-subtile image
+  __shared__ uint16_t lot[kSlots * 8];   // dO transpose image
+  __shared__ uint16_t lkt[kSlots * 8];   // K transpose image (fixed)
+  __shared__ uint16_t lst[kSlots * 8];   // dS^T transpose image (block)
+  __shared__ uint16_t lp[4][16 * kKT];   // per-wave P^T/dS^T A-image
+
+  const long bh = blockIdx.y;
+  const int j0 = blockIdx.x * kKT;       // key-tile origin
+  const long base = bh * (long)S * kDh;
+  const int lane = threadIdx.x % kWave;
+  const int wid = threadIdx.x / kWave;   // wave owns keys j0+wid*16..
+  const int l15 = lane & 15, l4 = lane >> 4;
+
+  // K,V register fragments (A-operand map: lane = key row l&15,
+  // dh chunk (l>>4)*8 + kk*32) for the wave's 16 keys
+  const int krow_g = j0 + wid * 16 + l15;
+  bf16x8 ka[2], va[2];
+#pragma unroll
+  for (int kk = 0; kk < 2; ++kk) {
+    const long off = base + (long)krow_g * kDh + kk * 32 + l4 * 8;
+    ka[kk] = *reinterpret_cast<const bf16x8*>(k + off);
+    va[kk] = *reinterpret_cast<const bf16x8*>(v + off);
+  }
+
+  // K transpose image (A-fragments of K^T for the dQ^T GEMM)
+#pragma unroll
+  for (int u = 0; u < kSlots / 256; ++u) {
+    const int f = u * 256 + threadIdx.x;
+    int m, c;
+    attn_slot_to_mc(f, m, c);
+    const uint16_t* src = k + base + (long)(j0 + m) * kDh + c;
+    __builtin_amdgcn_global_load_lds(
+        (glb_u32_t*)src, (lds_u32_t*)(lkt + (long)f * 8), 16, 0, 0);
+  }
+
+  f32x4 dk_acc[4], dv_acc[4];  // [dh col-block], rows = wave's 16 keys
+#pragma unroll
+  for (int db = 0; db < 4; ++db) {
+    dk_acc[db] = f32x4{0.f, 0.f, 0.f, 0.f};
+    dv_acc[db] = f32x4{0.f, 0.f, 0.f, 0.f};
+  }
+  __builtin_amdgcn_s_waitcnt(0);
+  __syncthreads();
+
+  uint16_t* lpw = lp[wid];
+  for (int i0 = 0; i0 < S; i0 += kKT) {
+    // ---- stage Q and dO in both images -----------------------------
+#pragma unroll
+    for (int u = 0; u < kSlots / 256; ++u) {
+      const int f = u * 256 + threadIdx.x;
+      const int kc = f >> 6, qr = f & 63;
+      int m, c;
+      attn_slot_to_mc(f, m, c);
+      const uint16_t* sq = q + base + (long)(i0 + qr) * kDh + kc * 8;
+      const uint16_t* sqt = q + base + (long)(i0 + m) * kDh + c;
+      const uint16_t* so = dout + base + (long)(i0 + qr) * kDh + kc * 8;
+      const uint16_t* sot = dout + base + (long)(i0 + m) * kDh + c;
+      __builtin_amdgcn_global_load_lds(
+          (glb_u32_t*)sq, (lds_u32_t*)(lq + (long)f * 8), 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (glb_u32_t*)sqt, (lds_u32_t*)(lqt + (long)f * 8), 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (glb_u32_t*)so, (lds_u32_t*)(lo + (long)f * 8), 16, 0, 0);
+      __builtin_amdgcn_global_load_lds(
+          (glb_u32_t*)sot, (lds_u32_t*)(lot + (long)f * 8), 16, 0, 0);
+    }
+    __builtin_amdgcn_s_waitcnt(0);
+    __syncthreads();
+
+    // ---- S^T = K Q^T (C-map: row = key (l4*4+r), col = qrow l15) ---
+    f32x4 st[4];
+#pragma unroll
+    for (int qc = 0; qc < 4; ++qc) {
+      st[qc] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        bf16x8 qf = *reinterpret_cast<const bf16x8*>(
+            lq + (((kk << 2) + l4) * 64 + (qc << 4) + l15) * 8);
+        st[qc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ka[kk], qf,
+                                                         st[qc], 0, 0, 0);
+      }
+    }
+
+    // ---- P^T = exp(S^T*scale - lse[qrow]); dP^T = V dO^T -----------
+    float p_val[4][4], lse_c[4], del_c[4];
+    f32x4 dp[4];
+#pragma unroll
+    for (int qc = 0; qc < 4; ++qc) {
+      lse_c[qc] = lse[bh * S + i0 + (qc << 4) + l15];
+      del_c[qc] = delta[bh * S + i0 + (qc << 4) + l15];
+      dp[qc] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        bf16x8 of = *reinterpret_cast<const bf16x8*>(
+            lo + (((kk << 2) + l4) * 64 + (qc << 4) + l15) * 8);
+        dp[qc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(va[kk], of,
+                                                         dp[qc], 0, 0, 0);
+      }
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        p_val[qc][r] = __expf(st[qc][r] * scale - lse_c[qc]);
+    }
+
+    // ---- dV += P^T dO (P^T via the per-wave A-image) ---------------
+#pragma unroll
+    for (int qc = 0; qc < 4; ++qc) {
+      const int so = ((qc << 1) + (l15 >> 3)) * 128 + (l15 & 7);
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        lpw[so + (((l4 << 2) + r) << 3)] = f2bf(p_val[qc][r]);
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_wave_barrier();
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 pa = *reinterpret_cast<const bf16x8*>(
+          lpw + (((kk << 2) + l4) * 16 + l15) * 8);
+#pragma unroll
+      for (int db = 0; db < 4; ++db) {
+        bf16x4a lo_ = attn_tr_read(lot, attn_frag_off(kk, db, 0, lane));
+        bf16x4a hi_ = attn_tr_read(lot, attn_frag_off(kk, db, 1, lane));
+        bf16x8 of =
+            __builtin_shufflevector(lo_, hi_, 0, 1, 2, 3, 4, 5, 6, 7);
+        dv_acc[db] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            pa, of, dv_acc[db], 0, 0, 0);
+      }
+    }
+
+    // ---- dS^T = P^T * (dP^T - delta[qrow]) * scale ------------------
+    // write into the per-wave A-image (for dK) AND the block transpose
+    // image (for dQ^T)
+#pragma unroll
+    for (int qc = 0; qc < 4; ++qc) {
+      const int so = ((qc << 1) + (l15 >> 3)) * 128 + (l15 & 7);
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const float ds =
+            p_val[qc][r] * (dp[qc][r] - del_c[qc]) * scale;
+        const uint16_t dsb = f2bf(ds);
+        lpw[so + (((l4 << 2) + r) << 3)] = dsb;
+        // transpose image element (m = key local, c = qrow)
+        const int m = (wid << 4) + (l4 << 2) + r;
+        const int c = (qc << 4) + l15;
+        const int blk = (((m >> 5) << 2) | (c >> 4)) * 8 +
+                        (((m >> 2) & 1) << 2) + ((m >> 3) & 3);
+        lst[blk * 64 + (m & 3) * 16 + (c & 15)] = dsb;
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    __builtin_amdgcn_wave_barrier();
+
+    // ---- dK += dS^T Q (Q via transpose image) ----------------------
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8 dsa = *reinterpret_cast<const bf16x8*>(
+          lpw + (((kk << 2) + l4) * 16 + l15) * 8);
+#pragma unroll
+      for (int db = 0; db < 4; ++db) {
+        bf16x4a lo_ = attn_tr_read(lqt, attn_frag_off(kk, db, 0, lane));
+        bf16x4a hi_ = attn_tr_read(lqt, attn_frag_off(kk, db, 1, lane));
+        bf16x8 qf =
+            __builtin_shufflevector(lo_, hi_, 0, 1, 2, 3, 4, 5, 6, 7);
+        dk_acc[db] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            dsa, qf, dk_acc[db], 0, 0, 0);
+      }
+    }
+    __syncthreads();  // all waves' dS^T visible in lst
+
+    // ---- dQ^T += K^T dS^T : wave w computes dh rows [w*16, w*16+16)
+    f32x4 dqt[4];
+#pragma unroll
+    for (int qc = 0; qc < 4; ++qc) {
+      dqt[qc] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+        bf16x4a klo = attn_tr_read(lkt, attn_frag_off(kk, wid, 0, lane));
+        bf16x4a khi = attn_tr_read(lkt, attn_frag_off(kk, wid, 1, lane));
+        bf16x8 ktf =
+            __builtin_shufflevector(klo, khi, 0, 1, 2, 3, 4, 5, 6, 7);
+        bf16x4a slo = attn_tr_read(lst, attn_frag_off(kk, qc, 0, lane));
+        bf16x4a shi = attn_tr_read(lst, attn_frag_off(kk, qc, 1, lane));
+        bf16x8 stf =
+            __builtin_shufflevector(slo, shi, 0, 1, 2, 3, 4, 5, 6, 7);
+        dqt[qc] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(ktf, stf,
+                                                          dqt[qc], 0, 0,
+                                                          0);
+      }
+      // accumulate transposed into dq[qrow][dh] (fp32 atomics)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const long qrow = i0 + (qc << 4) + l15;
+        const int dh = (wid << 4) + (l4 << 2) + r;
+        unsafeAtomicAdd(&dq_f32[(base + qrow * kDh) + dh], dqt[qc][r]);
+      }
+    }
+    __syncthreads();  // before the next i-tile overwrites LDS
+  }
+
+  // ---- store dK, dV (plain: this block owns keys j0+wid*16..) -------
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const long row = j0 + wid * 16 + (l4 << 2) + r;
+#pragma unroll
+    for (int db = 0; db < 4; ++db) {
+      dk_out[base + row * kDh + (db << 4) + l15] = f2bf(dk_acc[db][r]);
+      dv_out[base + row * kDh + (db << 4) + l15] = f2bf(dv_acc[db][r]);
+    }
+  }
+}
+
 }  // namespace
 
 // q,k,v: [B,H,S,64] bf16 contiguous -> (o [B,H,S,64] bf16, lse [B,H,S] f32)
@@ -247,6 +483,38 @@ std::tuple<at::Tensor, at::Tensor> attn_fwd_impl(const at::Tensor& q_in,
       reinterpret_cast<uint16_t*>(o.data_ptr()), lse.data_ptr<float>(), S,
       (float)scale);
   return {o, lse};
+}
+
+// grad,q,k,v,o: [B,H,S,64] bf16; lse: [B,H,S] f32 -> (dq, dk, dv) bf16
+std::tuple<at::Tensor, at::Tensor, at::Tensor> attn_bwd_impl(
+    const at::Tensor& grad_in, const at::Tensor& q_in,
+    const at::Tensor& k_in, const at::Tensor& v_in, const at::Tensor& o_in,
+    const at::Tensor& lse, double scale) {
+  auto g = grad_in.contiguous(), q = q_in.contiguous(),
+       k = k_in.contiguous(), v = v_in.contiguous(), o = o_in.contiguous();
+  TORCH_CHECK(q.scalar_type() == at::kBFloat16, "attn bwd: bf16 only");
+  TORCH_CHECK(q.dim() == 4 && q.size(3) == kDh && q.size(2) % 64 == 0,
+              "attn bwd: [B,H,S%64,64] expected");
+  const int B = q.size(0), H = q.size(1), S = q.size(2);
+  // delta_i = rowsum(dO_i * O_i) — one memory-bound pass
+  auto delta = (g.to(at::kFloat) * o.to(at::kFloat)).sum(-1);
+  auto dk = at::empty_like(k);
+  auto dv = at::empty_like(v);
+  auto dq_f32 = at::zeros({(long)B * H * S * kDh},
+                          q.options().dtype(at::kFloat));
+  auto s = at::cuda::getCurrentCUDAStream();
+  dim3 grid((unsigned)(S / kKT), (unsigned)(B * H));
+  attn_bwd_kernel<<<grid, 256, 0, s>>>(
+      reinterpret_cast<const uint16_t*>(q.data_ptr()),
+      reinterpret_cast<const uint16_t*>(k.data_ptr()),
+      reinterpret_cast<const uint16_t*>(v.data_ptr()),
+      reinterpret_cast<const uint16_t*>(g.data_ptr()),
+      lse.data_ptr<float>(), delta.data_ptr<float>(),
+      reinterpret_cast<uint16_t*>(dk.data_ptr()),
+      reinterpret_cast<uint16_t*>(dv.data_ptr()), dq_f32.data_ptr<float>(),
+      S, (float)scale);
+  auto dq = dq_f32.to(at::kBFloat16).view({B, H, S, (long)kDh});
+  return {dq, dk, dv};
 }
 
 }  // namespace dlx

@@ -42,6 +42,10 @@ std::tuple<at::Tensor, at::Tensor> attn_fwd_impl(const at::Tensor& q,
                                                  const at::Tensor& k,
                                                  const at::Tensor& v,
                                                  double scale);
+std::tuple<at::Tensor, at::Tensor, at::Tensor> attn_bwd_impl(
+    const at::Tensor& grad, const at::Tensor& q, const at::Tensor& k,
+    const at::Tensor& v, const at::Tensor& o, const at::Tensor& lse,
+    double scale);
 
 // batchnorm.hip
 std::tuple<at::Tensor, at::Tensor, at::Tensor, at::Tensor>
@@ -297,6 +301,7 @@ TORCH_LIBRARY(dlx_hip, m) {
   m.def("gelu_fwd(Tensor x) -> Tensor");
   m.def("gelu_bwd(Tensor g, Tensor x) -> Tensor");
   m.def("attn_fwd(Tensor q, Tensor k, Tensor v, float scale) -> (Tensor, Tensor)");
+  m.def("attn_bwd(Tensor grad, Tensor q, Tensor k, Tensor v, Tensor o, Tensor lse, float scale) -> (Tensor, Tensor, Tensor)");
 }
 
 TORCH_LIBRARY_IMPL(dlx_hip, CUDA, m) {
@@ -338,4 +343,5 @@ TORCH_LIBRARY_IMPL(dlx_hip, CUDA, m) {
   m.impl("gelu_fwd", dlx::gelu_fwd_impl);
   m.impl("gelu_bwd", dlx::gelu_bwd_impl);
   m.impl("attn_fwd", dlx::attn_fwd_impl);
+  m.impl("attn_bwd", dlx::attn_bwd_impl);
 }

@@ -487,20 +487,17 @@ class _AttentionMFMA(torch.autograd.Function):
         q, k, v, o, lse = ctx.saved_tensors
         scale = ctx.scale
         grad = grad.contiguous()
-        # recompute P row-exactly via the saved log-sum-exp; GEMMs stay
-        # bf16 (rocBLAS MFMA path — an all-fp32 recompute measured ~4%
-        # slower end-to-end than the bmm composition), softmax/delta
-        # arithmetic stays fp32.
-        s = (q @ k.transpose(-1, -2)).float() * scale    # [B,H,S,S]
-        p = torch.exp(s - lse.unsqueeze(-1))
-        pb = p.to(q.dtype)
-        dv = pb.transpose(-1, -2) @ grad
-        dp = (grad @ v.transpose(-1, -2)).float()
-        # dS = P * (dP - rowsum(dO * O))  (flash-bwd 'delta' trick)
-        delta = (grad.float() * o.float()).sum(-1, keepdim=True)
-        ds = (p * (dp - delta) * scale).to(q.dtype)
-        dq = ds @ k
-        dk = ds.transpose(-1, -2) @ q
+        import os
+        if os.environ.get("DLX_ATTN_BWD") == "compose":
+            # recompute-composition fallback (bf16 GEMMs, fp32 softmax)
+            s = (q @ k.transpose(-1, -2)).float() * scale
+            p = torch.exp(s - lse.unsqueeze(-1))
+            dv = p.to(q.dtype).transpose(-1, -2) @ grad
+            dp = (grad @ v.transpose(-1, -2)).float()
+            delta = (grad.float() * o.float()).sum(-1, keepdim=True)
+            ds = (p * (dp - delta) * scale).to(q.dtype)
+            return ds @ k, ds.transpose(-1, -2) @ q, dv, None
+        dq, dk, dv = native().attn_bwd(grad, q, k, v, o, lse, scale)
         return dq, dk, dv, None
 
 
