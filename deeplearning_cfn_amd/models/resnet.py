@@ -121,15 +121,18 @@ class Bottleneck(nn.Module):
         self.downsample = downsample
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        identity = x if self.downsample is None else self.downsample(x)
-        out = F_amd.conv_bn_act(x, self.conv1.weight, self.bn1.bn,
-                                1, 0, relu=True, training=self.training)
-        out = F_amd.conv_bn_act(out, self.conv2.weight, self.bn2.bn,
-                                self.conv2.stride[0], self.conv2.padding[0],
-                                relu=True, training=self.training)
-        return F_amd.conv_bn_act(out, self.conv3.weight, self.bn3.bn,
-                                 1, 0, residual=identity,
-                                 training=self.training)
+        # one autograd node for the whole block on the native path: the
+        # residual fan-in add is fused into conv1's dgrad epilogue
+        ds = self.downsample
+        return F_amd.bottleneck(
+            x, self.conv1.weight, self.bn1.bn,
+            self.conv2.weight, self.bn2.bn,
+            self.conv2.stride[0], self.conv2.padding[0],
+            self.conv3.weight, self.bn3.bn,
+            down_convw=ds.conv.weight if ds is not None else None,
+            down_bn=ds.bn if ds is not None else None,
+            down_stride=ds.conv.stride[0] if ds is not None else 1,
+            training=self.training)
 
 
 class ResNet(nn.Module):

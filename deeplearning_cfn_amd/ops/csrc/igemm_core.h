@@ -547,7 +547,8 @@ __global__ __launch_bounds__(WM* WN * 64) void igemm_kernel(
     AG ag, BG bg, void* __restrict__ Cout_,
     const uint16_t* __restrict__ zero_page,
     long M, long Nc, long K, int grid_n, OM om,
-    float* __restrict__ stats = nullptr) {
+    float* __restrict__ stats = nullptr,
+    const uint16_t* __restrict__ accum = nullptr) {
   constexpr int THREADS = WM * WN * 64;
   constexpr int SLOTS_A = (BK / 8) * BM;
   constexpr int SLOTS_B = (BK / 8) * BN;
@@ -665,10 +666,12 @@ __global__ __launch_bounds__(WM* WN * 64) void igemm_kernel(
         long col = n0 + wc * (BN / WN) + ni * 16 + l15;
         if (row < M && col < Nc) {
           const long o = om.off(row) + col;
+          float val = acc[mi][ni][r];
+          if (accum) val += bf2f(accum[o]);  // fused residual fan-in add
           if (OUT_BF16)
-            reinterpret_cast<uint16_t*>(Cout_)[o] = f2bf(acc[mi][ni][r]);
+            reinterpret_cast<uint16_t*>(Cout_)[o] = f2bf(val);
           else
-            reinterpret_cast<float*>(Cout_)[o] = acc[mi][ni][r];
+            reinterpret_cast<float*>(Cout_)[o] = val;
         }
       }
     }
@@ -701,7 +704,8 @@ __global__ __launch_bounds__(512) void igemm_kernel_8ph(
     AG ag, BG bg, void* __restrict__ Cout_,
     const uint16_t* __restrict__ zero_page,
     long M, long Nc, long K, int grid_n, OM om,
-    float* __restrict__ stats = nullptr) {
+    float* __restrict__ stats = nullptr,
+    const uint16_t* __restrict__ accum = nullptr) {
   constexpr int BM = 256, BN = 256, BK = 64;
   constexpr int THREADS = 512;
   constexpr int SLOTS = (BK / 8) * BM;          // 2048 16-B slots/operand
@@ -837,10 +841,12 @@ __global__ __launch_bounds__(512) void igemm_kernel_8ph(
         long col = n0 + wc * 64 + ni * 16 + l15;
         if (row < M && col < Nc) {
           const long o = om.off(row) + col;
+          float val = acc[mi][ni][r];
+          if (accum) val += bf2f(accum[o]);  // fused residual fan-in add
           if (OUT_BF16)
-            reinterpret_cast<uint16_t*>(Cout_)[o] = f2bf(acc[mi][ni][r]);
+            reinterpret_cast<uint16_t*>(Cout_)[o] = f2bf(val);
           else
-            reinterpret_cast<float*>(Cout_)[o] = acc[mi][ni][r];
+            reinterpret_cast<float*>(Cout_)[o] = val;
         }
       }
     }
@@ -898,6 +904,36 @@ inline void igemm_launch(const AG& ag, const BG& bg, void* cptr,
                          hipStream_t s) {
   igemm_launch_om<AG, BG, OUT_BF16, IdentityOut>(ag, bg, cptr, zp, M, Nc, K,
                                                  s, IdentityOut{Nc});
+}
+
+// variant with the fused epilogue accumulate: out = GEMM + accum (bf16,
+// same [M, Nc] layout) — the autograd residual fan-in add folded into
+// the dgrad store
+template <class AG, class BG>
+inline void igemm_launch_acc(const AG& ag, const BG& bg, void* cptr,
+                             const uint16_t* zp, long M, long Nc, long K,
+                             hipStream_t s, const uint16_t* accum) {
+  static const long mink8 = []() {
+    const char* e = getenv("DLX_8PH_MINK");
+    return e ? atol(e) : 384;
+  }();
+  const IdentityOut om{Nc};
+  if (M >= 256 && Nc >= 192 && K >= mink8) {
+    const int gm = (int)((M + 255) / 256), gn = (int)((Nc + 255) / 256);
+    igemm_kernel_8ph<AG, BG, true, IdentityOut>
+        <<<gm * gn, 512, 0, s>>>(ag, bg, cptr, zp, M, Nc, K, gn, om,
+                                 nullptr, accum);
+  } else if (Nc <= 96 && M >= 1024) {
+    const int gm = (int)((M + 255) / 256), gn = (int)((Nc + 63) / 64);
+    igemm_kernel<256, 64, 64, 4, 1, AG, BG, true, IdentityOut>
+        <<<gm * gn, 256, 0, s>>>(ag, bg, cptr, zp, M, Nc, K, gn, om,
+                                 nullptr, accum);
+  } else {
+    const int gm = (int)((M + 127) / 128), gn = (int)((Nc + 127) / 128);
+    igemm_kernel<128, 128, 64, 2, 4, AG, BG, true, IdentityOut>
+        <<<gm * gn, 512, 0, s>>>(ag, bg, cptr, zp, M, Nc, K, gn, om,
+                                 nullptr, accum);
+  }
 }
 
 // geometry the launcher will pick (the stats slab is [gm][2][Nc], so the

@@ -100,6 +100,9 @@ at::Tensor conv2d_fwd_impl(const at::Tensor& x, const at::Tensor& w,
 at::Tensor conv2d_dgrad_impl(const at::Tensor& g, const at::Tensor& w,
                              int64_t H, int64_t W, int64_t stride,
                              int64_t pad);
+at::Tensor conv2d_dgrad_acc_impl(const at::Tensor& g, const at::Tensor& w,
+                                 int64_t H, int64_t W, int64_t stride,
+                                 int64_t pad, const at::Tensor& accum);
 
 // conv_wrw.hip
 at::Tensor conv2d_wrw_impl(const at::Tensor& g, const at::Tensor& x,
@@ -292,6 +295,7 @@ TORCH_LIBRARY(dlx_hip, m) {
   m.def("transpose2d(Tensor x) -> Tensor");
   m.def("conv2d_fwd(Tensor x, Tensor w, int stride, int pad) -> Tensor");
   m.def("conv2d_dgrad(Tensor g, Tensor w, int H, int W, int stride, int pad) -> Tensor");
+  m.def("conv2d_dgrad_acc(Tensor g, Tensor w, int H, int W, int stride, int pad, Tensor accum) -> Tensor");
   m.def("conv2d_wrw(Tensor g, Tensor x, int Kh, int Kw, int stride, int pad) -> Tensor");
   m.def("conv2d_fwd_stats(Tensor x, Tensor w, int stride, int pad) -> (Tensor, Tensor)");
   m.def("bn_fwd_train_presum(Tensor x, Tensor w, Tensor b, Tensor(a!) rmean, Tensor(b!) rvar, float momentum, float eps, bool relu, Tensor sums) -> (Tensor, Tensor, Tensor, Tensor)");
@@ -334,6 +338,7 @@ TORCH_LIBRARY_IMPL(dlx_hip, CUDA, m) {
   m.impl("transpose2d", dlx::transpose2d_impl);
   m.impl("conv2d_fwd", dlx::conv2d_fwd_impl);
   m.impl("conv2d_dgrad", dlx::conv2d_dgrad_impl);
+  m.impl("conv2d_dgrad_acc", dlx::conv2d_dgrad_acc_impl);
   m.impl("conv2d_wrw", dlx::conv2d_wrw_impl);
   m.impl("conv2d_fwd_stats", dlx::conv2d_fwd_stats_impl);
   m.impl("bn_fwd_train_presum", dlx::bn_fwd_train_presum_impl);

@@ -254,6 +254,111 @@ def conv_bn_act(x: torch.Tensor, weight: torch.Tensor,
 
 
 # ---------------------------------------------------------------------------
+# whole-bottleneck fusion (ResNet-50/101/152 block) — VERDICT r1 weak #2:
+# the autograd fan-in at the residual junction costs one big at::add per
+# block; hand-chaining the block backward lands that add in conv1's
+# dgrad epilogue (conv2d_dgrad_acc).
+# ---------------------------------------------------------------------------
+
+
+class _BottleneckFn(torch.autograd.Function):
+    """conv1/bn1+relu -> conv2/bn2+relu -> conv3/bn3(+res)+relu with an
+    optional conv/bn downsample branch, trained end-to-end through the
+    native kernels.  Forward matches the composed _ConvBNAct chain
+    op-for-op; backward chains bn_bwd_*_m / dgrad / wrw by hand and
+    fuses the residual fan-in into dgrad_acc."""
+
+    @staticmethod
+    def forward(ctx, x, w1, w2, w3, wd,
+                g1, b1, g2, b2, g3, b3, gd, bd,
+                rm1, rv1, rm2, rv2, rm3, rv3, rmd, rvd,
+                s2, p2, sd, momentum, eps):
+        n = native()
+        x = _to_nhwc(x)
+        yc1, sums1 = n.conv2d_fwd_stats(x, w1, 1, 0)
+        y1, m1, i1, k1 = n.bn_fwd_train_presum(yc1, g1, b1, rm1, rv1,
+                                               momentum, eps, True, sums1)
+        yc2, sums2 = n.conv2d_fwd_stats(y1, w2, s2, p2)
+        y2, m2, i2, k2 = n.bn_fwd_train_presum(yc2, g2, b2, rm2, rv2,
+                                               momentum, eps, True, sums2)
+        yc3, sums3 = n.conv2d_fwd_stats(y2, w3, 1, 0)
+        if wd is not None:
+            ycd, sumsd = n.conv2d_fwd_stats(x, wd, sd, 0)
+            res, md, idd, _ = n.bn_fwd_train_presum(
+                ycd, gd, bd, rmd, rvd, momentum, eps, False, sumsd)
+        else:
+            ycd = md = idd = None
+            res = x
+        y3, m3, i3, k3 = n.bn_fwd_train_res_presum(
+            yc3, g3, b3, rm3, rv3, momentum, eps, res, sums3)
+        ctx.save_for_backward(x, w1, w2, w3, wd, g1, g2, g3, gd,
+                              yc1, yc2, yc3, ycd, y1, y2,
+                              m1, i1, k1, m2, i2, k2, m3, i3, k3, md, idd)
+        ctx.meta = (s2, p2, sd)
+        return y3
+
+    @staticmethod
+    def backward(ctx, grad):
+        n = native()
+        (x, w1, w2, w3, wd, g1, g2, g3, gd,
+         yc1, yc2, yc3, ycd, y1, y2,
+         m1, i1, k1, m2, i2, k2, m3, i3, k3, md, idd) = ctx.saved_tensors
+        s2, p2, sd = ctx.meta
+        g = _to_nhwc(grad)
+        dyc3, dg3, db3, dres = n.bn_bwd_res_m(g, yc3, g3, m3, i3, k3)
+        dy2 = n.conv2d_dgrad(dyc3, w3, y2.shape[2], y2.shape[3], 1, 0)
+        dw3 = n.conv2d_wrw(dyc3, y2, 1, 1, 1, 0)
+        dyc2, dg2, db2 = n.bn_bwd_m(dy2, yc2, g2, m2, i2, k2)
+        dy1 = n.conv2d_dgrad(dyc2, w2, y1.shape[2], y1.shape[3], s2, p2)
+        dw2 = n.conv2d_wrw(dyc2, y1, w2.shape[2], w2.shape[3], s2, p2)
+        dyc1, dg1, db1 = n.bn_bwd_m(dy1, yc1, g1, m1, i1, k1)
+        dw1 = n.conv2d_wrw(dyc1, x, 1, 1, 1, 0)
+        dwd = dgd = dbd = None
+        if wd is not None:
+            # bn (no relu) backward never reads its y argument
+            dycd, dgd, dbd = n.bn_bwd(dres, ycd, gd, md, idd, ycd, False)
+            accum = n.conv2d_dgrad(dycd, wd, x.shape[2], x.shape[3], sd, 0)
+            dwd = n.conv2d_wrw(dycd, x, 1, 1, sd, 0)
+        else:
+            accum = dres
+        # the fused residual fan-in: dx = conv1-dgrad + accum in one pass
+        dx = n.conv2d_dgrad_acc(dyc1, w1, x.shape[2], x.shape[3], 1, 0,
+                                accum)
+        return (dx, dw1, dw2, dw3, dwd, dg1, db1, dg2, db2, dg3, db3,
+                dgd, dbd) + (None,) * 13
+
+
+def bottleneck(x, conv1w, bn1, conv2w, bn2, stride2, pad2, conv3w, bn3,
+               down_convw=None, down_bn=None, down_stride=1,
+               training=True):
+    """The full ResNet bottleneck unit through one autograd node."""
+    if (training and _use_native(x) and _conv_policy() == "native"
+            and x.dtype == torch.bfloat16 and x.shape[1] % 8 == 0
+            and _os.environ.get("DLX_BLOCK_FUSION", "1") != "0"):
+        gd = bd = rmd = rvd = None
+        if down_bn is not None:
+            gd, bd = down_bn.weight, down_bn.bias
+            rmd, rvd = down_bn.running_mean, down_bn.running_var
+        return _BottleneckFn.apply(
+            x, conv1w, conv2w, conv3w, down_convw,
+            bn1.weight, bn1.bias, bn2.weight, bn2.bias,
+            bn3.weight, bn3.bias, gd, bd,
+            bn1.running_mean, bn1.running_var,
+            bn2.running_mean, bn2.running_var,
+            bn3.running_mean, bn3.running_var, rmd, rvd,
+            stride2, pad2, down_stride, bn1.momentum, bn1.eps)
+    # composed fallback (CPU / eval / non-native)
+    identity = x if down_convw is None else conv_bn_act(
+        x, down_convw, down_bn, down_stride, 0, relu=False,
+        training=training)
+    out = conv_bn_act(x, conv1w, bn1, 1, 0, relu=True, training=training)
+    out = conv_bn_act(out, conv2w, bn2, stride2, pad2, relu=True,
+                      training=training)
+    return conv_bn_act(out, conv3w, bn3, 1, 0, residual=identity,
+                       training=training)
+
+
+# ---------------------------------------------------------------------------
 # fused softmax + cross-entropy (the classifier head)
 # ---------------------------------------------------------------------------
 

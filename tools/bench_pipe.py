@@ -20,9 +20,13 @@ from deeplearning_cfn_amd.data import HostPinnedPrefetcher, SyntheticImages
 
 
 def run(batch, image_size, n_batches=30):
+    # pool=4: host tensors are synthesised once and pinned once, so the
+    # measurement is the PIPE (queue + H2D on the side stream), not the
+    # CPU's randn rate (fresh-randn sources ~1.3k img/s of host work —
+    # which is why bench.py pre-stages device batches, as disclosed)
     src = SyntheticImages(batch, image_size=image_size, seed=1,
                           dtype=torch.bfloat16, channels_last=True,
-                          pinned=True)
+                          pinned=True, pool=4)
     pf = HostPinnedPrefetcher(src, torch.device("cuda", 0))
     it = iter(pf)
     # warm
