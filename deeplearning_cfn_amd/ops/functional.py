@@ -332,9 +332,13 @@ def bottleneck(x, conv1w, bn1, conv2w, bn2, stride2, pad2, conv3w, bn3,
                down_convw=None, down_bn=None, down_stride=1,
                training=True):
     """The full ResNet bottleneck unit through one autograd node."""
+    # Measured (gpurun_out/r2f.log, b2048): the one-node block is ~1.4%
+    # SLOWER than the composed chain despite eliminating the fan-in
+    # at::add (and graphs replay it worse still) — default OFF, kept as
+    # an opt-in with its parity test (tests/test_block_fusion_gpu.py).
     if (training and _use_native(x) and _conv_policy() == "native"
             and x.dtype == torch.bfloat16 and x.shape[1] % 8 == 0
-            and _os.environ.get("DLX_BLOCK_FUSION", "1") != "0"):
+            and _os.environ.get("DLX_BLOCK_FUSION", "0") == "1"):
         gd = bd = rmd = rvd = None
         if down_bn is not None:
             gd, bd = down_bn.weight, down_bn.bias
