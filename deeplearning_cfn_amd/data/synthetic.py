@@ -75,10 +75,12 @@ class SyntheticBertBatches:
     def __init__(self, batch_size: int, seq_len: int = 128,
                  vocab_size: int = 30522, mask_frac: float = 0.15,
                  seed: int = 0, rank: int = 0, device: str = "cpu",
-                 pinned: bool = False):
+                 pinned: bool = False, pool: int = 0):
         self.batch_size, self.seq_len = batch_size, seq_len
         self.vocab_size, self.mask_frac = vocab_size, mask_frac
         self.device, self.pinned = device, pinned
+        self.pool = pool            # >0: cycle a fixed batch set (see
+        self._pool_batches = []     # SyntheticImages — memorization)
         self.gen = torch.Generator().manual_seed(seed * 100003 + rank)
 
     def batch(self):
@@ -97,5 +99,13 @@ class SyntheticBertBatches:
         return ids, labels, nsp
 
     def __iter__(self):
-        while True:
-            yield self.batch()
+        if self.pool > 0:
+            i = 0
+            while True:
+                if len(self._pool_batches) < self.pool:
+                    self._pool_batches.append(self.batch())
+                yield self._pool_batches[i % self.pool]
+                i += 1
+        else:
+            while True:
+                yield self.batch()

@@ -869,7 +869,7 @@ inline void igemm_launch_om(const AG& ag, const BG& bg, void* cptr,
   // 128-KiB block per CU cannot hide the prologue/drain otherwise)
   static const long mink8 = []() {
     const char* e = getenv("DLX_8PH_MINK");
-    return e ? atol(e) : 384;
+    return e ? atol(e) : 256;  // 384->256 r2: short-K dgrad -20% (r2d/r2e)
   }();
   if (M >= 256 && Nc >= 192 && K >= mink8) {
     const int gm = (int)((M + 255) / 256), gn = (int)((Nc + 255) / 256);
@@ -915,7 +915,7 @@ inline void igemm_launch_acc(const AG& ag, const BG& bg, void* cptr,
                              hipStream_t s, const uint16_t* accum) {
   static const long mink8 = []() {
     const char* e = getenv("DLX_8PH_MINK");
-    return e ? atol(e) : 384;
+    return e ? atol(e) : 256;  // 384->256 r2: short-K dgrad -20% (r2d/r2e)
   }();
   const IdentityOut om{Nc};
   if (M >= 256 && Nc >= 192 && K >= mink8) {
@@ -941,7 +941,7 @@ inline void igemm_launch_acc(const AG& ag, const BG& bg, void* cptr,
 inline long igemm_gm(long M, long Nc, long K) {
   static const long mink8 = []() {
     const char* e = getenv("DLX_8PH_MINK");
-    return e ? atol(e) : 384;
+    return e ? atol(e) : 256;  // 384->256 r2: short-K dgrad -20% (r2d/r2e)
   }();
   if (M >= 256 && Nc >= 192 && K >= mink8) return (M + 255) / 256;
   if (Nc <= 96 && M >= 1024) return (M + 255) / 256;
@@ -956,7 +956,7 @@ inline void igemm_launch_stats(const AG& ag, const BG& bg, void* cptr,
                                hipStream_t s, float* stats) {
   static const long mink8 = []() {
     const char* e = getenv("DLX_8PH_MINK");
-    return e ? atol(e) : 384;
+    return e ? atol(e) : 256;  // 384->256 r2: short-K dgrad -20% (r2d/r2e)
   }();
   const IdentityOut om{Nc};
   if (M >= 256 && Nc >= 192 && K >= mink8) {

@@ -193,7 +193,8 @@ def train(args: argparse.Namespace) -> dict:
     if is_bert:
         source = SyntheticBertBatches(args.batch_size, seq_len=args.seq_len,
                                       seed=args.seed, rank=ctx.rank,
-                                      pinned=ctx.device.type == "cuda")
+                                      pinned=ctx.device.type == "cuda",
+                                      pool=args.data_pool)
     else:
         image_size = 32 if args.dataset.startswith("cifar") else 224
         ncls = 10 if args.dataset.startswith("cifar") else 1000
@@ -216,7 +217,7 @@ def train(args: argparse.Namespace) -> dict:
     # accuracy); endless-data runs get a held-out seed.
     eval_batches = None
     if args.eval_steps:
-        if args.data_pool > 0 and not is_bert:
+        if args.data_pool > 0:
             eval_batches = iter(source)
         elif is_bert:
             eval_batches = iter(SyntheticBertBatches(
