@@ -128,8 +128,13 @@ __global__ __launch_bounds__(WM* WN * 64) void conv_wrw_kernel(
     int xcd = bid % 8, idx = bid / 8;
     bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
   }
-  // tile-major: a tile's slices stay adjacent (same XCD chunk)
-  const int tile = bid / split, slice = bid % split;
+  // slice-major: the TILES of one m-slice are adjacent in remapped-bid
+  // space, i.e. co-resident on ONE XCD — they read the SAME g/x m-rows,
+  // so the slice's operand traffic is fetched into that XCD's L2 once
+  // instead of once per tile (the re-staging amplification is
+  // (Rows/BM + Cols/BN)-fold otherwise).
+  const int ntiles = nwg / split;
+  const int slice = bid / ntiles, tile = bid % ntiles;
   const int bm = tile / grid_n, bn = tile % grid_n;
   const long row0 = (long)bm * kBMw, col0 = (long)bn * kBNw;
   const long mbeg = slice * m_per_slice;
