@@ -6,10 +6,13 @@ equivalent for a launch-bound inner loop is HIP graph capture (guide:
 captures {zero_grad, forward, loss, backward, finalize, optimizer} once
 against static input buffers and replays it per step.
 
-Scope: world_size == 1 by default.  RCCL collectives during capture are
-allowed when ``capture_collectives=True`` (torch's NCCL/RCCL process
-group supports capture on ROCm); the multi-GPU default stays eager until
-that path is validated on an 8-GPU box.
+Scope: world_size == 1 by default.  RCCL collectives DO capture and
+replay correctly on this stack — proven on hardware at world 1
+(tests/test_rccl_gpu.py::test_nccl_ws1_init_and_collectives captures an
+all_reduce and replays it) — but the multi-GPU default stays eager
+(``DLX_GRAPHS=1`` to force) until an 8-GPU box can validate multi-rank
+capture; at the bench's batch sizes the step is not launch-bound, so
+eager costs little there.
 """
 
 from __future__ import annotations
