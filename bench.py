@@ -34,10 +34,10 @@ def parse_args():
     p.add_argument("--gpus", type=int, default=1)
     p.add_argument("--steps", type=int, default=30)
     p.add_argument("--warmup", type=int, default=10)
-    p.add_argument("--batch-size", type=int, default=2048,
-                   help="per-GPU batch (weak scaling; 2048 = ~130 GB of "
+    p.add_argument("--batch-size", type=int, default=2560,
+                   help="per-GPU batch (weak scaling; 2560 = ~163 GB of "
                         "the 288 GB HBM3E per GPU — measured fastest: "
-                        "profiles/r50_batch_sweep.log)")
+                        "9 246 vs 9 169 img/s at 2048, gpurun_out/r2p.log)")
     p.add_argument("--num-layers", type=int, default=50)
     p.add_argument("--dtype", default="bf16")
     p.add_argument("--impl", default="auto")
