@@ -242,8 +242,10 @@ __global__ __launch_bounds__(WM* WN * 64) void conv_wrw_kernel(
 // ---------------------------------------------------------------------
 // 256x256 8-phase wrw (Co >= 192): the igemm_kernel_8ph pipeline
 // (counted vmcnt, raw barriers, one LDS object, stage units = (operand,
-// k-half)) with tr-read fragments and the atomic fp32 epilogue.  Halves
-// the staged bytes per MAC vs the 128^2 tile — wrw is staging-BW-bound.
+// k-half)) with tr-read fragments.  Halves the staged bytes per MAC vs
+// the 128^2 tile — wrw is staging-BW-bound.  Round-2 retrofit: hoisted
+// prep()/at() gathers, slice-major XCD scheduling, deterministic
+// partial-plane epilogue (plain stores; host sums).
 // ---------------------------------------------------------------------
 
 template <class AG, class BG>
@@ -265,7 +267,9 @@ __global__ __launch_bounds__(512) void conv_wrw_kernel_8ph(
     int xcd = bid % 8, idx = bid / 8;
     bid = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + idx;
   }
-  const int tile = bid / split, slice = bid % split;
+  // slice-major: one m-slice's tiles co-resident on one XCD's L2
+  const int ntiles = nwg / split;
+  const int slice = bid / ntiles, tile = bid % ntiles;
   const int bm = tile / grid_n, bn = tile % grid_n;
   const long row0 = (long)bm * BMw, col0 = (long)bn * BNw;
   const long mbeg = slice * m_per_slice;
@@ -280,22 +284,34 @@ __global__ __launch_bounds__(512) void conv_wrw_kernel_8ph(
   f32x4 acc[MR][NR] = {};
   const long ntile = (mend - mbeg + BKw - 1) / BKw;
 
+  // hoisted per-slot state: 4 (half, i) slots per thread per operand
+  int mloc8[4];
+  typename AG::Pre apre8[4];
+  typename BG::Pre bpre8[4];
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    const int f = j * THREADS + threadIdx.x;
+    int c;
+    slot_to_mc<4>(f, mloc8[j], c);
+    apre8[j] = ag.prep(mloc8[j], row0 + c);
+    bpre8[j] = bg.prep(mloc8[j], col0 + c);
+  }
+
   // stage unit u of tile t: u=0 A-kstep0, u=1 B-kstep0, u=2 A-ks1, u=3 B-ks1
   auto stage_unit = [&](int buf, long t, int u) {
     const long mt0 = mbeg + t * BKw;
+    const long mlim = mend - mt0;
     const bool isB = u & 1;
     const int half = u >> 1;
     uint16_t* l0 = lds + (buf * 2 + (isB ? 1 : 0)) * SLOTS * 8;
 #pragma unroll
     for (int i = 0; i < 2; ++i) {
-      const int f = (half * 2 + i) * THREADS + threadIdx.x;
-      int mloc, c;
-      slot_to_mc<4>(f, mloc, c);
-      const long m = mt0 + mloc;
+      const int j = half * 2 + i;
+      const int f = j * THREADS + threadIdx.x;
+      const bool mv = mloc8[j] < mlim;
       const uint16_t* src =
-          (m < mend) ? (isB ? bg.src(m, col0 + c, zp)
-                            : ag.src(m, row0 + c, zp))
-                     : zp;
+          isB ? bg.at(bpre8[j], mt0, mloc8[j], mv, zp)
+              : ag.at(apre8[j], mt0, mloc8[j], mv, zp);
       __builtin_amdgcn_global_load_lds(
           (glb_u32_t*)src, (lds_u32_t*)(l0 + (long)f * 8), 16, 0, 0);
     }
@@ -363,6 +379,8 @@ __global__ __launch_bounds__(512) void conv_wrw_kernel_8ph(
   phase_mfma(buf, 2);
   phase_mfma(buf, 3);
 
+  // deterministic partial-plane epilogue (host sums the slices)
+  float* op = out + (long)slice * Rows * Cols;
 #pragma unroll
   for (int mi = 0; mi < MR; ++mi) {
 #pragma unroll
@@ -371,12 +389,8 @@ __global__ __launch_bounds__(512) void conv_wrw_kernel_8ph(
       for (int r = 0; r < 4; ++r) {
         const long row = row0 + wr * 128 + mi * 16 + l4 * 4 + r;
         const long col = col0 + wc * 64 + ni * 16 + l15;
-        if (row < Rows && col < Cols) {
-          if (split > 1)
-            unsafeAtomicAdd(&out[row * Cols + col], acc[mi][ni][r]);
-          else
-            out[row * Cols + col] = acc[mi][ni][r];
-        }
+        if (row < Rows && col < Cols)
+          op[row * Cols + col] = acc[mi][ni][r];
       }
     }
   }
@@ -465,14 +479,17 @@ at::Tensor conv2d_wrw_impl(const at::Tensor& g_in, const at::Tensor& x_in,
   if (p.Ci <= 4) return conv2d_wrw_stem(g, x, p);
 
   const long Rows = p.Co, Cols = p.K;
-  // Measured (profiles/): the 128^2 4-wave tile beats the 256^2 8-phase
-  // variant on every ResNet wrw shape (occupancy/tail balance outweighs
-  // the halved staged bytes) — keep 8ph behind an env flag for re-tests.
-  static const bool want8ph = []() {
+  // r2 A/B (gpurun_out/r2q.log, with hoisted gathers + slice-major +
+  // partial planes): the 256^2 8-phase tile wins +5-10% on Rows>=512 &&
+  // Cols>=512 shapes (halved staged bytes per MAC) and loses on 256-row
+  // ones (occupancy) — per-shape policy, env-overridable:
+  // DLX_WRW_8PH=0 never, 1 = the old >=192 gate, default >=512.
+  static const int mode8 = []() {
     const char* e = getenv("DLX_WRW_8PH");
-    return e && e[0] == '1';
+    return e ? atoi(e) : 2;
   }();
-  const bool big = want8ph && Rows >= 192 && Cols >= 192;
+  const bool big = (mode8 == 1 && Rows >= 192 && Cols >= 192) ||
+                   (mode8 == 2 && Rows >= 512 && Cols >= 512);
   const int BM = big ? 256 : kBMw, BN = big ? 256 : kBNw;
   const int gm = (int)((Rows + BM - 1) / BM);
   const int gn = (int)((Cols + BN - 1) / BN);
@@ -501,21 +518,29 @@ at::Tensor conv2d_wrw_impl(const at::Tensor& g_in, const at::Tensor& x_in,
     return !(e && e[0] == '4');
   }();
 
-  at::Tensor outf;
+  // deterministic split-m: one partial plane per slice, summed below
+  at::Tensor outf =
+      at::empty({split, Rows, Cols}, x.options().dtype(at::kFloat));
   if (big) {
-    // 8-phase experiment path keeps the atomic-accumulate epilogue
-    outf = (split > 1)
-               ? at::zeros({Rows, Cols}, x.options().dtype(at::kFloat))
-               : at::empty({Rows, Cols}, x.options().dtype(at::kFloat));
-    ConvFwdGather bg{reinterpret_cast<const uint16_t*>(x.data_ptr()), p};
-    conv_wrw_kernel_8ph<RowGather, ConvFwdGather>
-        <<<nblocks, 512, 0, s>>>(
-            ag, bg, outf.data_ptr<float>(),
-            reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, Rows,
-            Cols, gn, (int)split, m_per_slice);
+    // 256^2 8-phase pipeline (half the staged bytes per MAC); r2
+    // retrofit shares the hoisted gathers + partial planes
+    const bool plain = p.Kh == 1 && p.Kw == 1 && p.stride == 1 && p.pad == 0;
+    if (plain) {
+      RowGather bg{reinterpret_cast<const uint16_t*>(x.data_ptr()), p.M,
+                   (long)p.Ci};
+      conv_wrw_kernel_8ph<RowGather, RowGather><<<nblocks, 512, 0, s>>>(
+          ag, bg, outf.data_ptr<float>(),
+          reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, Rows,
+          Cols, gn, (int)split, m_per_slice);
+    } else {
+      WrwXGather bg{reinterpret_cast<const uint16_t*>(x.data_ptr()), p};
+      conv_wrw_kernel_8ph<RowGather, WrwXGather><<<nblocks, 512, 0, s>>>(
+          ag, bg, outf.data_ptr<float>(),
+          reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, Rows,
+          Cols, gn, (int)split, m_per_slice);
+    }
+    if (split > 1) outf = outf.sum(0);
   } else {
-    // deterministic split-m: one partial plane per slice, summed below
-    outf = at::empty({split, Rows, Cols}, x.options().dtype(at::kFloat));
     const bool plain = p.Kh == 1 && p.Kw == 1 && p.stride == 1 && p.pad == 0;
     if (plain) {
       // 1x1/s1: x~ IS x ([M, Ci] row-major) — no im2col address math
