@@ -110,16 +110,25 @@ DLX_DEV int frag_off(int kk, int cb, int i, int lane) {
 //    stores) summed by the host — deterministic (no fp32 atomics, the
 //    ADVICE r1 determinism item) and free of atomic contention on
 //    small outputs (64x64 dW had 768 blocks hammering 16 KB).
-template <class AG, class BG, int WM = 2, int WN = 2>
+// BM/BN in {64,128}: the host picks 64-wide operand tiles when the
+// output extent is <=64 (layer-1-class dW), halving wasted zero-page
+// staging and doubling occupancy there.
+template <class AG, class BG, int WM = 2, int WN = 2, int BM = 128,
+          int BN = 128>
 __global__ __launch_bounds__(WM* WN * 64) void conv_wrw_kernel(
     AG ag, BG bg, float* __restrict__ out,  // [split][Rows][Cols]
     const uint16_t* __restrict__ zero_page, long Mred, long Rows, long Cols,
     int grid_n, int split, long m_per_slice) {
   constexpr int THREADS = WM * WN * 64;
-  constexpr int MR = (128 / WM) / 16;
-  constexpr int NR = (128 / WN) / 16;
-  constexpr int SPT = kSlotsW / THREADS;  // staging slots per thread
-  __shared__ uint16_t lds[2 * 2 * kSlotsW * 8];  // [buf][A|B][slots][8]
+  constexpr int MR = (BM / WM) / 16;
+  constexpr int NR = (BN / WN) / 16;
+  constexpr int LCBA = (BM == 128) ? 3 : 2;   // log2(col-blocks)
+  constexpr int LCBB = (BN == 128) ? 3 : 2;
+  constexpr int SLOTS_A = kBKw * BM / 8;
+  constexpr int SLOTS_B = kBKw * BN / 8;
+  constexpr int SPA = (SLOTS_A + THREADS - 1) / THREADS;
+  constexpr int SPB = (SLOTS_B + THREADS - 1) / THREADS;
+  __shared__ uint16_t lds[2 * (SLOTS_A + SLOTS_B) * 8];  // [buf][A|B]
 
   int nwg = gridDim.x;
   int bid = blockIdx.x;
@@ -136,7 +145,7 @@ __global__ __launch_bounds__(WM* WN * 64) void conv_wrw_kernel(
   const int ntiles = nwg / split;
   const int slice = bid / ntiles, tile = bid % ntiles;
   const int bm = tile / grid_n, bn = tile % grid_n;
-  const long row0 = (long)bm * kBMw, col0 = (long)bn * kBNw;
+  const long row0 = (long)bm * BM, col0 = (long)bn * BN;
   const long mbeg = slice * m_per_slice;
   const long mend = (mbeg + m_per_slice < Mred) ? mbeg + m_per_slice : Mred;
   if (mbeg >= mend) return;
@@ -150,31 +159,44 @@ __global__ __launch_bounds__(WM* WN * 64) void conv_wrw_kernel(
   const long ntile = (mend - mbeg + kBKw - 1) / kBKw;
 
   // hoist per-slot loop invariants (slot id is fixed per thread)
-  int mloc_[SPT];
-  typename AG::Pre apre[SPT];
-  typename BG::Pre bpre[SPT];
+  int mla[SPA], mlb[SPB];
+  typename AG::Pre apre[SPA];
+  typename BG::Pre bpre[SPB];
 #pragma unroll
-  for (int u = 0; u < SPT; ++u) {
+  for (int u = 0; u < SPA; ++u) {
     const int f = u * THREADS + threadIdx.x;
     int c;
-    slot_to_mc<3>(f, mloc_[u], c);
-    apre[u] = ag.prep(mloc_[u], row0 + c);
-    bpre[u] = bg.prep(mloc_[u], col0 + c);
+    slot_to_mc<LCBA>(f, mla[u], c);
+    if (f < SLOTS_A) apre[u] = ag.prep(mla[u], row0 + c);
+  }
+#pragma unroll
+  for (int u = 0; u < SPB; ++u) {
+    const int f = u * THREADS + threadIdx.x;
+    int c;
+    slot_to_mc<LCBB>(f, mlb[u], c);
+    if (f < SLOTS_B) bpre[u] = bg.prep(mlb[u], col0 + c);
   }
 
   auto stage = [&](int buf, long t) {
     const long mt0 = mbeg + t * kBKw;
     const long mlim = mend - mt0;
-    uint16_t* la = lds + (buf * 2 + 0) * kSlotsW * 8;
-    uint16_t* lb = lds + (buf * 2 + 1) * kSlotsW * 8;
+    uint16_t* la = lds + buf * (SLOTS_A + SLOTS_B) * 8;
+    uint16_t* lb = la + SLOTS_A * 8;
 #pragma unroll
-    for (int u = 0; u < SPT; ++u) {
+    for (int u = 0; u < SPA; ++u) {
       const int f = u * THREADS + threadIdx.x;
-      const bool mv = mloc_[u] < mlim;
-      const uint16_t* sa = ag.at(apre[u], mt0, mloc_[u], mv, zero_page);
-      const uint16_t* sb = bg.at(bpre[u], mt0, mloc_[u], mv, zero_page);
+      if (f >= SLOTS_A) break;
+      const bool mv = mla[u] < mlim;
+      const uint16_t* sa = ag.at(apre[u], mt0, mla[u], mv, zero_page);
       __builtin_amdgcn_global_load_lds(
           (glb_u32_t*)sa, (lds_u32_t*)(la + (long)f * 8), 16, 0, 0);
+    }
+#pragma unroll
+    for (int u = 0; u < SPB; ++u) {
+      const int f = u * THREADS + threadIdx.x;
+      if (f >= SLOTS_B) break;
+      const bool mv = mlb[u] < mlim;
+      const uint16_t* sb = bg.at(bpre[u], mt0, mlb[u], mv, zero_page);
       __builtin_amdgcn_global_load_lds(
           (glb_u32_t*)sb, (lds_u32_t*)(lb + (long)f * 8), 16, 0, 0);
     }
@@ -187,8 +209,8 @@ __global__ __launch_bounds__(WM* WN * 64) void conv_wrw_kernel(
   int buf = 0;
   for (long t = 0; t < ntile; ++t) {
     if (t + 1 < ntile) stage(buf ^ 1, t + 1);
-    const uint16_t* la = lds + (buf * 2 + 0) * kSlotsW * 8;
-    const uint16_t* lb = lds + (buf * 2 + 1) * kSlotsW * 8;
+    const uint16_t* la = lds + buf * (SLOTS_A + SLOTS_B) * 8;
+    const uint16_t* lb = la + SLOTS_A * 8;
 #pragma unroll
     for (int kk = 0; kk < kBKw / 32; ++kk) {
       // A fragments: rows (co) of this wave's strip
@@ -196,15 +218,15 @@ __global__ __launch_bounds__(WM* WN * 64) void conv_wrw_kernel(
 #pragma unroll
       for (int mi = 0; mi < MR; ++mi) {
         const int cb = wr * MR + mi;
-        bf16x4 lo = tr_read(la, frag_off<3>(kk, cb, 0, lane));
-        bf16x4 hi = tr_read(la, frag_off<3>(kk, cb, 1, lane));
+        bf16x4 lo = tr_read(la, frag_off<LCBA>(kk, cb, 0, lane));
+        bf16x4 hi = tr_read(la, frag_off<LCBA>(kk, cb, 1, lane));
         af[mi] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
       }
 #pragma unroll
       for (int ni = 0; ni < NR; ++ni) {
         const int cb = wc * NR + ni;
-        bf16x4 lo = tr_read(lb, frag_off<3>(kk, cb, 0, lane));
-        bf16x4 hi = tr_read(lb, frag_off<3>(kk, cb, 1, lane));
+        bf16x4 lo = tr_read(lb, frag_off<LCBB>(kk, cb, 0, lane));
+        bf16x4 hi = tr_read(lb, frag_off<LCBB>(kk, cb, 1, lane));
         bfr[ni] = __builtin_shufflevector(lo, hi, 0, 1, 2, 3, 4, 5, 6, 7);
       }
       __builtin_amdgcn_s_setprio(1);
@@ -230,8 +252,8 @@ __global__ __launch_bounds__(WM* WN * 64) void conv_wrw_kernel(
     for (int ni = 0; ni < NR; ++ni) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const long row = row0 + wr * (128 / WM) + mi * 16 + l4 * 4 + r;
-        const long col = col0 + wc * (128 / WN) + ni * 16 + l15;
+        const long row = row0 + wr * (BM / WM) + mi * 16 + l4 * 4 + r;
+        const long col = col0 + wc * (BN / WN) + ni * 16 + l15;
         if (row < Rows && col < Cols)
           op[row * Cols + col] = acc[mi][ni][r];
       }
@@ -403,6 +425,32 @@ at::Tensor stem_pack_x(const at::Tensor& x_cl, int Wp, int pad);
 
 namespace {
 
+// 2-phase wrw launch with per-shape tile selection: 64-wide operand
+// tiles when the output extent is <=64 (no zero-page half), 8-wave
+// 128^2 otherwise.
+template <class BG>
+void launch_wrw_2ph(const RowGather& ag, const BG& bg, float* outp,
+                    const uint16_t* zp, long M, long Rows, long Cols,
+                    int gm, int gn, int split, long mps, bool w8,
+                    hipStream_t s) {
+  const int nblocks = gm * gn * split;
+  if (Rows <= 64 && Cols <= 64)
+    conv_wrw_kernel<RowGather, BG, 2, 2, 64, 64><<<nblocks, 256, 0, s>>>(
+        ag, bg, outp, zp, M, Rows, Cols, gn, split, mps);
+  else if (Rows <= 64)
+    conv_wrw_kernel<RowGather, BG, 2, 2, 64, 128><<<nblocks, 256, 0, s>>>(
+        ag, bg, outp, zp, M, Rows, Cols, gn, split, mps);
+  else if (Cols <= 64)
+    conv_wrw_kernel<RowGather, BG, 2, 2, 128, 64><<<nblocks, 256, 0, s>>>(
+        ag, bg, outp, zp, M, Rows, Cols, gn, split, mps);
+  else if (w8)
+    conv_wrw_kernel<RowGather, BG, 2, 4><<<nblocks, 512, 0, s>>>(
+        ag, bg, outp, zp, M, Rows, Cols, gn, split, mps);
+  else
+    conv_wrw_kernel<RowGather, BG><<<nblocks, kThreadsW, 0, s>>>(
+        ag, bg, outp, zp, M, Rows, Cols, gn, split, mps);
+}
+
 // small-Cin stem weight grad: reduce-major GEMM over the packed x4p
 // (StemWrwGather, igemm_core.h); dW columns are (dh, dwp, c4) and the
 // dwp/channel pads are sliced away on the host.
@@ -439,11 +487,10 @@ at::Tensor conv2d_wrw_stem(const at::Tensor& g, const at::Tensor& x,
   auto s = at::cuda::getCurrentCUDAStream();
   RowGather ag{reinterpret_cast<const uint16_t*>(g.data_ptr()), p.M,
                (long)p.Co};
-  conv_wrw_kernel<RowGather, StemWrwGather, 2, 4>
-      <<<gm * gn * (int)split, 512, 0, s>>>(
-          ag, bg, outf.data_ptr<float>(),
-          reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, Rows,
-          Cols, gn, (int)split, m_per_slice);
+  launch_wrw_2ph(ag, bg, outf.data_ptr<float>(),
+                 reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M,
+                 Rows, Cols, gm, gn, (int)split, m_per_slice, /*w8=*/true,
+                 s);
   auto summed = (split > 1) ? outf.sum(0) : outf.view({Rows, Cols});
   // [Co, Kh, Kwp, 4] -> slice off the pads -> dW with CL strides
   return summed.to(at::kBFloat16)
@@ -546,28 +593,14 @@ at::Tensor conv2d_wrw_impl(const at::Tensor& g_in, const at::Tensor& x_in,
       // 1x1/s1: x~ IS x ([M, Ci] row-major) — no im2col address math
       RowGather bg{reinterpret_cast<const uint16_t*>(x.data_ptr()), p.M,
                    (long)p.Ci};
-      if (w8)
-        conv_wrw_kernel<RowGather, RowGather, 2, 4><<<nblocks, 512, 0, s>>>(
-            ag, bg, outf.data_ptr<float>(),
-            reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, Rows,
-            Cols, gn, (int)split, m_per_slice);
-      else
-        conv_wrw_kernel<RowGather, RowGather><<<nblocks, kThreadsW, 0, s>>>(
-            ag, bg, outf.data_ptr<float>(),
-            reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, Rows,
-            Cols, gn, (int)split, m_per_slice);
+      launch_wrw_2ph(ag, bg, outf.data_ptr<float>(),
+                     reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M,
+                     Rows, Cols, gm, gn, (int)split, m_per_slice, w8, s);
     } else {
       WrwXGather bg{reinterpret_cast<const uint16_t*>(x.data_ptr()), p};
-      if (w8)
-        conv_wrw_kernel<RowGather, WrwXGather, 2, 4><<<nblocks, 512, 0, s>>>(
-            ag, bg, outf.data_ptr<float>(),
-            reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, Rows,
-            Cols, gn, (int)split, m_per_slice);
-      else
-        conv_wrw_kernel<RowGather, WrwXGather><<<nblocks, kThreadsW, 0, s>>>(
-            ag, bg, outf.data_ptr<float>(),
-            reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M, Rows,
-            Cols, gn, (int)split, m_per_slice);
+      launch_wrw_2ph(ag, bg, outf.data_ptr<float>(),
+                     reinterpret_cast<const uint16_t*>(zp.data_ptr()), p.M,
+                     Rows, Cols, gm, gn, (int)split, m_per_slice, w8, s);
     }
     if (split > 1) outf = outf.sum(0);
   }
@@ -606,11 +639,9 @@ at::Tensor gemm_tn_mred_impl(const at::Tensor& a_in, const at::Tensor& b_in) {
   auto s = at::cuda::getCurrentCUDAStream();
   RowGather ag{reinterpret_cast<const uint16_t*>(A.data_ptr()), M, Rows};
   RowGather bg{reinterpret_cast<const uint16_t*>(B.data_ptr()), M, Cols};
-  conv_wrw_kernel<RowGather, RowGather>
-      <<<gm * gn * (int)split, kThreadsW, 0, s>>>(
-          ag, bg, outf.data_ptr<float>(),
-          reinterpret_cast<const uint16_t*>(zp.data_ptr()), M, Rows, Cols,
-          gn, (int)split, m_per_slice);
+  launch_wrw_2ph(ag, bg, outf.data_ptr<float>(),
+                 reinterpret_cast<const uint16_t*>(zp.data_ptr()), M, Rows,
+                 Cols, gm, gn, (int)split, m_per_slice, /*w8=*/false, s);
   if (split > 1) outf = outf.sum(0);
   return outf.to(at::kBFloat16).view({Rows, Cols});
 }
