@@ -37,7 +37,7 @@ def parse_args():
     p.add_argument("--batch-size", type=int, default=2560,
                    help="per-GPU batch (weak scaling; 2560 = ~163 GB of "
                         "the 288 GB HBM3E per GPU — measured fastest: "
-                        "9 246 vs 9 169 img/s at 2048, gpurun_out/r2p.log)")
+                        "9 246 vs 9 169 img/s at 2048, profiles/r2_experiments/r2p.log)")
     p.add_argument("--num-layers", type=int, default=50)
     p.add_argument("--dtype", default="bf16")
     p.add_argument("--impl", default="auto")

@@ -526,7 +526,7 @@ at::Tensor conv2d_wrw_impl(const at::Tensor& g_in, const at::Tensor& x_in,
   if (p.Ci <= 4) return conv2d_wrw_stem(g, x, p);
 
   const long Rows = p.Co, Cols = p.K;
-  // r2 A/B (gpurun_out/r2q.log, with hoisted gathers + slice-major +
+  // r2 A/B (profiles/r2_experiments/r2q.log, with hoisted gathers + slice-major +
   // partial planes): the 256^2 8-phase tile wins +5-10% on Rows>=512 &&
   // Cols>=512 shapes (halved staged bytes per MAC) and loses on 256-row
   // ones (occupancy) — per-shape policy, env-overridable:

@@ -332,7 +332,7 @@ def bottleneck(x, conv1w, bn1, conv2w, bn2, stride2, pad2, conv3w, bn3,
                down_convw=None, down_bn=None, down_stride=1,
                training=True):
     """The full ResNet bottleneck unit through one autograd node."""
-    # Measured (gpurun_out/r2f.log, b2048): the one-node block is ~1.4%
+    # Measured (profiles/r2_experiments/r2f.log, b2048): the one-node block is ~1.4%
     # SLOWER than the composed chain despite eliminating the fan-in
     # at::add (and graphs replay it worse still) — default OFF, kept as
     # an opt-in with its parity test (tests/test_block_fusion_gpu.py).
